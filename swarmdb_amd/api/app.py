@@ -1,0 +1,619 @@
+"""FastAPI server — the REST compatibility surface.
+
+All 19 reference endpoints (reference api.py:365-945, inventoried in
+SURVEY.md §2.3) plus the probes the reference left unrouted
+(get_agent_load, unread count, LLM dispatch). Reference defects fixed
+without breaking documented behavior:
+
+- /messages/broadcast and /groups/message declare the dict response shape
+  they actually return;
+- no ``status`` name shadowing in handlers;
+- the app module is real (``swarmdb_amd.api.app:app``).
+
+Auth model kept from the reference: any non-empty username/password gets a
+token (api.py:373-380, documented demo behavior — SURVEY.md §8.10, behind
+a validator hook), and ``sub == "admin"`` is the admin check.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import threading
+import time
+from datetime import datetime, timedelta, timezone
+from typing import Callable, Dict, List, Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Query, Request, status as http
+from fastapi.middleware.cors import CORSMiddleware
+from fastapi.responses import JSONResponse
+from fastapi.security import HTTPAuthorizationCredentials, HTTPBearer
+
+from ..core.config import QueueConfig
+from ..core.message import MessagePriority, MessageStatus, MessageType
+from ..runtime.facade import SwarmsDB
+from ..utils import jwt as jwtlib
+from .models import (
+    AgentGroupRequest,
+    AgentLoadResponse,
+    AgentRegistrationRequest,
+    BroadcastRequest,
+    BroadcastResponse,
+    GroupMessageRequest,
+    GroupMessageResponse,
+    HealthResponse,
+    LLMDispatchResponse,
+    MessageRequest,
+    MessageResponse,
+    SystemStats,
+    Token,
+    UserCredentials,
+)
+
+logger = logging.getLogger("swarmdb_amd.api")
+
+API_VERSION = "1.0.0"
+
+
+class ApiSettings:
+    """Env-var config tier (reference api.py:38-52)."""
+
+    def __init__(self) -> None:
+        env = os.environ
+        self.api_env = env.get("API_ENV", "development")
+        self.jwt_secret = env.get("JWT_SECRET", "supersecretkey")
+        self.jwt_algorithm = env.get("JWT_ALGORITHM", "HS256")
+        self.token_expire_minutes = int(env.get("TOKEN_EXPIRE_MINUTES", "1440"))
+        self.rate_limit_per_minute = int(env.get("RATE_LIMIT_PER_MINUTE", "300"))
+        self.cors_origins = env.get("CORS_ORIGINS", "*").split(",")
+        self.port = int(env.get("PORT", "8000"))
+
+
+class RateLimiter:
+    """Per-client-IP sliding 60 s window (reference api.py:266-314), with
+    a bounded table (stale IPs evicted — the reference's dict grew without
+    bound) and a lock (the reference mutated it from 4 threads unlocked,
+    SURVEY.md §5.2)."""
+
+    def __init__(self, limit_per_minute: int):
+        self.limit = limit_per_minute
+        self._hits: Dict[str, List[float]] = {}
+        self._lock = threading.Lock()
+
+    def allow(self, client_ip: str) -> bool:
+        now = time.time()
+        with self._lock:
+            window = self._hits.setdefault(client_ip, [])
+            cutoff = now - 60.0
+            while window and window[0] < cutoff:
+                window.pop(0)
+            if len(window) >= self.limit:
+                return False
+            window.append(now)
+            if len(self._hits) > 10000:
+                for ip in [ip for ip, w in self._hits.items()
+                           if not w or w[-1] < cutoff]:
+                    del self._hits[ip]
+            return True
+
+
+def create_app(
+    db: Optional[SwarmsDB] = None,
+    settings: Optional[ApiSettings] = None,
+    credential_validator: Optional[Callable[[str, str], bool]] = None,
+) -> FastAPI:
+    """App factory. ``db`` defaults to a SwarmsDB built from env config
+    (GPU engine if a device is visible, CPU otherwise). One shared
+    SwarmsDB per process — the device-owner model
+    replaces the reference's per-worker divergent state (§8.8)."""
+    settings = settings or ApiSettings()
+    if db is None:
+        db = SwarmsDB(config=QueueConfig.from_env())
+
+    from contextlib import asynccontextmanager
+
+    @asynccontextmanager
+    async def lifespan(app_: FastAPI):
+        yield
+        # reference api.py:939-945 (shutdown hook)
+        db.close()
+
+    app = FastAPI(
+        title="SwarmDB (MI355X)",
+        description="GPU-resident agent message queue and LLM load balancer",
+        version=API_VERSION,
+        lifespan=lifespan,
+    )
+    app.state.db = db
+    app.state.settings = settings
+
+    app.add_middleware(
+        CORSMiddleware,
+        allow_origins=settings.cors_origins,
+        allow_credentials=True,
+        allow_methods=["*"],
+        allow_headers=["*"],
+    )
+
+    limiter = RateLimiter(settings.rate_limit_per_minute)
+
+    @app.middleware("http")
+    async def rate_limit(request: Request, call_next):
+        client_ip = request.client.host if request.client else "unknown"
+        if not limiter.allow(client_ip):
+            return JSONResponse(
+                status_code=http.HTTP_429_TOO_MANY_REQUESTS,
+                content={"detail": "Rate limit exceeded"},
+            )
+        return await call_next(request)
+
+    bearer = HTTPBearer(auto_error=False)
+
+    def create_access_token(username: str) -> str:
+        """reference api.py:318-334"""
+        expire = datetime.now(timezone.utc) + timedelta(
+            minutes=settings.token_expire_minutes
+        )
+        return jwtlib.encode(
+            {"sub": username, "exp": expire.timestamp()},
+            settings.jwt_secret,
+            settings.jwt_algorithm,
+        )
+
+    def get_current_agent(
+        creds: Optional[HTTPAuthorizationCredentials] = Depends(bearer),
+    ) -> str:
+        """reference api.py:337-361"""
+        if creds is None:
+            raise HTTPException(
+                status_code=http.HTTP_401_UNAUTHORIZED,
+                detail="Not authenticated",
+                headers={"WWW-Authenticate": "Bearer"},
+            )
+        try:
+            payload = jwtlib.decode(
+                creds.credentials, settings.jwt_secret, settings.jwt_algorithm
+            )
+        except jwtlib.JWTError as e:
+            raise HTTPException(
+                status_code=http.HTTP_401_UNAUTHORIZED,
+                detail=f"Invalid token: {e}",
+                headers={"WWW-Authenticate": "Bearer"},
+            ) from None
+        sub = payload.get("sub")
+        if not sub:
+            raise HTTPException(
+                status_code=http.HTTP_401_UNAUTHORIZED, detail="Invalid token"
+            )
+        return sub
+
+    def is_admin(agent: str) -> bool:
+        # authorization model: sub == "admin" (reference api.py:398 etc.)
+        return agent == "admin"
+
+    # ---------------- auth ----------------
+
+    @app.post("/auth/token", response_model=Token)
+    async def login(credentials: UserCredentials):
+        """reference api.py:365-388 — accepts any non-empty credentials
+        (documented demo behavior) unless a validator
+        hook was installed."""
+        if not credentials.username or not credentials.password:
+            raise HTTPException(
+                status_code=http.HTTP_400_BAD_REQUEST,
+                detail="Username and password required",
+            )
+        if credential_validator is not None and not credential_validator(
+            credentials.username, credentials.password
+        ):
+            raise HTTPException(
+                status_code=http.HTTP_401_UNAUTHORIZED,
+                detail="Invalid credentials",
+            )
+        return Token(access_token=create_access_token(credentials.username))
+
+    # ---------------- agents ----------------
+
+    @app.post("/agents/register", status_code=http.HTTP_201_CREATED)
+    async def register_agent(
+        req: AgentRegistrationRequest, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:391-437 (self-or-admin)."""
+        if req.agent_id != current and not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="Can only register yourself unless admin",
+            )
+        db.register_agent(req.agent_id)
+        meta = {}
+        if req.description:
+            meta["description"] = req.description
+        if req.capabilities:
+            meta["capabilities"] = req.capabilities
+        if req.metadata:
+            meta.update(req.metadata)
+        if meta:
+            db.agent_metadata[req.agent_id] = meta
+        return {"status": "registered", "agent_id": req.agent_id}
+
+    @app.delete("/agents/{agent_id}")
+    async def deregister_agent(
+        agent_id: str, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:440-469 (self-or-admin)."""
+        if agent_id != current and not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="Can only deregister yourself unless admin",
+            )
+        if not db.deregister_agent(agent_id):
+            raise HTTPException(
+                status_code=http.HTTP_404_NOT_FOUND,
+                detail=f"Agent {agent_id} not registered",
+            )
+        db.agent_metadata.pop(agent_id, None)
+        return {"status": "deregistered", "agent_id": agent_id}
+
+    # ---------------- messages ----------------
+
+    @app.post("/messages", response_model=MessageResponse)
+    async def send_message(
+        req: MessageRequest, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:472-504 — sender is the authenticated agent."""
+        try:
+            mid = db.send_message(
+                sender_id=current,
+                content=req.content,
+                receiver_id=req.receiver_id,
+                message_type=req.message_type,
+                priority=req.priority,
+                metadata=req.metadata,
+                visible_to=req.visible_to,
+            )
+        except Exception as e:
+            raise HTTPException(
+                status_code=http.HTTP_500_INTERNAL_SERVER_ERROR,
+                detail=f"Failed to send message: {e}",
+            ) from None
+        msg = db.get_message(mid)
+        return MessageResponse.from_message(msg)
+
+    @app.post("/messages/broadcast", response_model=BroadcastResponse)
+    async def broadcast_message(
+        req: BroadcastRequest, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:507-536; declared model fixed to the dict it
+        returns."""
+        mid = db.broadcast_message(
+            sender_id=current,
+            content=req.content,
+            message_type=req.message_type,
+            priority=req.priority,
+            metadata=req.metadata,
+            exclude_agents=req.exclude_agents,
+        )
+        return BroadcastResponse(status="broadcast", message_id=mid)
+
+    @app.get("/messages/{message_id}", response_model=MessageResponse)
+    async def get_message(
+        message_id: str, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:539-568 — visibility: admin, sender, receiver,
+        or listed in visible_to."""
+        msg = db.get_message(message_id)
+        if msg is None:
+            raise HTTPException(
+                status_code=http.HTTP_404_NOT_FOUND,
+                detail=f"Message {message_id} not found",
+            )
+        allowed = (
+            is_admin(current)
+            or msg.sender_id == current
+            or msg.receiver_id == current
+            or current in msg.visible_to
+        )
+        if not allowed:
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="Not authorized to view this message",
+            )
+        return MessageResponse.from_message(msg)
+
+    @app.get("/messages", response_model=List[MessageResponse])
+    async def query_messages(
+        sender_id: Optional[str] = Query(None),
+        receiver_id: Optional[str] = Query(None),
+        message_type: Optional[MessageType] = Query(None),
+        status: Optional[MessageStatus] = Query(None),
+        after_timestamp: Optional[float] = Query(None),
+        before_timestamp: Optional[float] = Query(None),
+        limit: int = Query(100, ge=1, le=1000),
+        current: str = Depends(get_current_agent),
+    ):
+        """reference api.py:571-621 — non-admins may only query their own
+        traffic (the reference's scope check at api.py:594-601); handler
+        keeps the query-param name ``status`` without shadowing the
+        status-codes module."""
+        if not is_admin(current):
+            if sender_id is None and receiver_id is None:
+                sender_id = current
+            elif sender_id != current and receiver_id != current:
+                raise HTTPException(
+                    status_code=http.HTTP_403_FORBIDDEN,
+                    detail="Can only query your own messages",
+                )
+        msgs = db.query_messages(
+            sender_id=sender_id,
+            receiver_id=receiver_id,
+            message_type=message_type,
+            status=status,
+            after_timestamp=after_timestamp,
+            before_timestamp=before_timestamp,
+            limit=limit,
+        )
+        return [MessageResponse.from_message(m) for m in msgs]
+
+    @app.get("/agents/{agent_id}/messages", response_model=List[MessageResponse])
+    async def get_agent_messages(
+        agent_id: str,
+        status: Optional[MessageStatus] = Query(None),
+        limit: int = Query(100, ge=1, le=1000),
+        skip: int = Query(0, ge=0),
+        current: str = Depends(get_current_agent),
+    ):
+        """reference api.py:624-664 (self-or-admin)."""
+        if agent_id != current and not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="Can only view your own messages",
+            )
+        msgs = db.get_agent_messages(agent_id, status=status, limit=limit, skip=skip)
+        return [MessageResponse.from_message(m) for m in msgs]
+
+    @app.post("/agents/receive", response_model=List[MessageResponse])
+    async def receive_messages(
+        max_messages: int = Query(100, ge=1, le=10000),
+        timeout: float = Query(1.0, ge=0.0, le=30.0),
+        priority_order: bool = Query(False),
+        current: str = Depends(get_current_agent),
+    ):
+        """The consumer-poll endpoint (reference api.py:667-688)."""
+        msgs = db.receive_messages(
+            current, max_messages=max_messages, timeout=timeout,
+            priority_order=priority_order,
+        )
+        return [MessageResponse.from_message(m) for m in msgs]
+
+    @app.put("/messages/{message_id}/status")
+    async def update_message_status(
+        message_id: str,
+        new_status: MessageStatus = Query(..., alias="status"),
+        current: str = Depends(get_current_agent),
+    ):
+        """reference api.py:691-733 (receiver-or-admin; 'processed' goes
+        through mark_message_as_processed)."""
+        msg = db.get_message(message_id)
+        if msg is None:
+            raise HTTPException(
+                status_code=http.HTTP_404_NOT_FOUND,
+                detail=f"Message {message_id} not found",
+            )
+        if msg.receiver_id != current and not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="Only the receiver can update message status",
+            )
+        if new_status == MessageStatus.PROCESSED:
+            db.mark_message_as_processed(message_id)
+        else:
+            db.update_message_status(message_id, new_status)
+        return {"status": "updated", "message_id": message_id,
+                "new_status": new_status.value}
+
+    # ---------------- groups ----------------
+
+    @app.post("/groups", status_code=http.HTTP_201_CREATED)
+    async def create_group(
+        req: AgentGroupRequest, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:736-757 — any authenticated agent."""
+        db.add_agent_group(req.group_name, req.agent_ids)
+        return {"status": "created", "group_name": req.group_name,
+                "agent_count": len(req.agent_ids)}
+
+    @app.post("/groups/message", response_model=GroupMessageResponse)
+    async def send_group_message(
+        req: GroupMessageRequest, current: str = Depends(get_current_agent)
+    ):
+        """reference api.py:760-787; declared model fixed."""
+        try:
+            ids = db.send_to_group(
+                group_name=req.group_name,
+                sender_id=current,
+                content=req.content,
+                message_type=req.message_type,
+                priority=req.priority,
+                metadata=req.metadata,
+            )
+        except ValueError as e:
+            raise HTTPException(
+                status_code=http.HTTP_404_NOT_FOUND, detail=str(e)
+            ) from None
+        return GroupMessageResponse(status="sent", message_ids=ids)
+
+    # ---------------- probes ----------------
+
+    @app.get("/health", response_model=HealthResponse)
+    async def health():
+        """reference api.py:790-815 — no auth; probes the engine instead
+        of the Kafka admin client."""
+        try:
+            total = db.engine.total_messages()
+            connected = True
+        except Exception:
+            total, connected = 0, False
+        return HealthResponse(
+            status="healthy" if connected else "degraded",
+            version=API_VERSION,
+            engine=type(db.engine).__name__,
+            engine_connected=connected,
+            registered_agents=len(db.registered_agents),
+            total_messages=total,
+        )
+
+    @app.get("/stats", response_model=SystemStats)
+    async def get_stats(current: str = Depends(get_current_agent)):
+        """reference api.py:818-838 (admin only)."""
+        if not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
+            )
+        return SystemStats(**db.get_stats())
+
+    @app.get("/agents/{agent_id}/load", response_model=AgentLoadResponse)
+    async def agent_load(
+        agent_id: str, current: str = Depends(get_current_agent)
+    ):
+        """get_agent_load, unrouted in the reference (swarmdb/
+        main.py:1049-1094) — self-or-admin."""
+        if agent_id != current and not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
+            )
+        return AgentLoadResponse(**db.get_agent_load(agent_id))
+
+    @app.get("/agents/{agent_id}/unread_count")
+    async def unread_count(
+        agent_id: str, current: str = Depends(get_current_agent)
+    ):
+        """get_unread_message_count, unrouted in the reference
+        (swarmdb/ main.py:1026-1047)."""
+        if agent_id != current and not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
+            )
+        return {"agent_id": agent_id,
+                "unread_count": db.get_unread_message_count(agent_id)}
+
+    # ---------------- LLM load balancing ----------------
+
+    @app.post("/llm/backends/{backend_id}")
+    async def register_backend(
+        backend_id: str, current: str = Depends(get_current_agent)
+    ):
+        if not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
+            )
+        idx = db.register_llm_backend(backend_id)
+        return {"status": "registered", "backend_id": backend_id, "index": idx}
+
+    @app.post("/llm/dispatch", response_model=LLMDispatchResponse)
+    async def dispatch(current: str = Depends(get_current_agent)):
+        """Least-loaded dispatch (BASELINE config 5) — the argmin
+        reduction kernel on GPU."""
+        try:
+            backend = db.dispatch_llm_request(current)
+        except RuntimeError as e:
+            raise HTTPException(
+                status_code=http.HTTP_503_SERVICE_UNAVAILABLE, detail=str(e)
+            ) from None
+        return LLMDispatchResponse(backend_id=backend)
+
+    @app.post("/llm/complete/{backend_id}")
+    async def complete(
+        backend_id: str, current: str = Depends(get_current_agent)
+    ):
+        db.complete_llm_request(backend_id)
+        return {"status": "completed", "backend_id": backend_id}
+
+    # ---------------- admin ----------------
+
+    def _require_admin(current: str) -> None:
+        if not is_admin(current):
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
+            )
+
+    @app.post("/admin/save")
+    async def admin_save(current: str = Depends(get_current_agent)):
+        """reference api.py:841-861."""
+        _require_admin(current)
+        path = db.save_message_history()
+        return {"status": "saved", "path": path}
+
+    @app.post("/admin/load")
+    async def admin_load(
+        path: str = Query(...), current: str = Depends(get_current_agent)
+    ):
+        """load_message_history over HTTP (never routed in the reference —
+        SURVEY.md §5.4)."""
+        _require_admin(current)
+        try:
+            n = db.load_message_history(path)
+        except FileNotFoundError:
+            raise HTTPException(
+                status_code=http.HTTP_404_NOT_FOUND,
+                detail=f"History file not found: {path}",
+            ) from None
+        return {"status": "loaded", "messages": n}
+
+    @app.post("/admin/flush")
+    async def admin_flush(
+        older_than: Optional[float] = Query(None),
+        current: str = Depends(get_current_agent),
+    ):
+        """reference api.py:864-885."""
+        _require_admin(current)
+        n = db.flush_old_messages(older_than)
+        return {"status": "flushed", "messages_flushed": n}
+
+    @app.post("/admin/resend_failed")
+    async def admin_resend(current: str = Depends(get_current_agent)):
+        """reference api.py:888-912."""
+        _require_admin(current)
+        ids = db.resend_failed_messages()
+        return {"status": "resent", "message_ids": ids}
+
+    @app.post("/admin/scale_partitions")
+    async def admin_scale(current: str = Depends(get_current_agent)):
+        """reference api.py:915-935 — the elasticity knob."""
+        _require_admin(current)
+        return db.auto_scale_partitions()
+
+    @app.post("/admin/export_yaml")
+    async def admin_export_yaml(current: str = Depends(get_current_agent)):
+        """export_as_yaml over HTTP (unrouted in the reference)."""
+        _require_admin(current)
+        return {"status": "exported", "path": db.export_as_yaml()}
+
+    return app
+
+
+# module-level app for `uvicorn swarmdb_amd.api.app:app` (the reference's
+# broken `app:app` reference fixed — SURVEY.md §8.6)
+app: Optional[FastAPI] = None
+
+
+def get_app() -> FastAPI:
+    global app
+    if app is None:
+        app = create_app()
+    return app
+
+
+def main() -> None:  # pragma: no cover
+    import uvicorn
+
+    settings = ApiSettings()
+    uvicorn.run(
+        create_app(settings=settings),
+        host="0.0.0.0",
+        port=settings.port,
+        log_level="info",
+    )
+
+
+if __name__ == "__main__":  # pragma: no cover
+    main()

@@ -1,0 +1,318 @@
+"""API-layer tests: all 19 reference endpoints + auth/permission matrix
+ against the CPU engine via FastAPI TestClient."""
+
+import json
+from pathlib import Path
+
+import pytest
+from fastapi.testclient import TestClient
+
+from swarmdb_amd import QueueConfig, SwarmsDB
+from swarmdb_amd.api.app import ApiSettings, create_app
+
+
+@pytest.fixture()
+def client(tmp_path):
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path / "hist"),
+                      max_agents=256)
+    db = SwarmsDB(config=cfg)
+    app = create_app(db=db, settings=ApiSettings())
+    with TestClient(app) as c:
+        c.db = db
+        yield c
+    db.config.auto_save = False
+
+
+def token(client, username):
+    r = client.post("/auth/token",
+                    json={"username": username, "password": "pw"})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["token_type"] == "bearer"
+    return body["access_token"]
+
+
+def auth(client, username):
+    return {"Authorization": f"Bearer {token(client, username)}"}
+
+
+def test_token_requires_nonempty_credentials(client):
+    r = client.post("/auth/token", json={"username": "", "password": "x"})
+    assert r.status_code == 400
+    r = client.post("/auth/token", json={"username": "a", "password": ""})
+    assert r.status_code == 400
+    # any non-empty pair accepted (reference api.py:373-380)
+    r = client.post("/auth/token", json={"username": "anyone", "password": "pw"})
+    assert r.status_code == 200
+
+
+def test_endpoints_require_auth(client):
+    for method, path in [
+        ("post", "/agents/register"),
+        ("delete", "/agents/x"),
+        ("post", "/messages"),
+        ("post", "/messages/broadcast"),
+        ("get", "/messages/xyz"),
+        ("get", "/messages"),
+        ("get", "/agents/x/messages"),
+        ("post", "/agents/receive"),
+        ("put", "/messages/x/status?status=read"),
+        ("post", "/groups"),
+        ("post", "/groups/message"),
+        ("get", "/stats"),
+        ("post", "/admin/save"),
+        ("post", "/admin/flush"),
+        ("post", "/admin/resend_failed"),
+        ("post", "/admin/scale_partitions"),
+    ]:
+        r = getattr(client, method)(path)
+        assert r.status_code == 401, f"{method} {path} -> {r.status_code}"
+
+
+def test_invalid_token_rejected(client):
+    r = client.get("/messages", headers={"Authorization": "Bearer garbage"})
+    assert r.status_code == 401
+
+
+def test_register_self_and_admin(client):
+    h = auth(client, "agent1")
+    r = client.post("/agents/register", headers=h,
+                    json={"agent_id": "agent1", "description": "test bot",
+                          "capabilities": ["chat"]})
+    assert r.status_code == 201
+    assert r.json() == {"status": "registered", "agent_id": "agent1"}
+    # non-admin cannot register someone else
+    r = client.post("/agents/register", headers=h, json={"agent_id": "other"})
+    assert r.status_code == 403
+    # admin can
+    r = client.post("/agents/register", headers=auth(client, "admin"),
+                    json={"agent_id": "other"})
+    assert r.status_code == 201
+    # metadata stashed
+    assert client.db.agent_metadata["agent1"]["description"] == "test bot"
+
+
+def test_deregister(client):
+    h = auth(client, "agent1")
+    client.post("/agents/register", headers=h, json={"agent_id": "agent1"})
+    r = client.delete("/agents/agent1", headers=h)
+    assert r.status_code == 200
+    r = client.delete("/agents/agent1", headers=h)
+    assert r.status_code == 404
+    r = client.delete("/agents/someone", headers=h)
+    assert r.status_code == 403
+
+
+def test_send_and_get_message(client):
+    h = auth(client, "alice")
+    r = client.post("/messages", headers=h,
+                    json={"receiver_id": "bob", "content": "hi bob",
+                          "message_type": "chat", "priority": 2})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["sender_id"] == "alice"
+    assert body["receiver_id"] == "bob"
+    assert body["content"] == "hi bob"
+    assert body["type"] == "chat"
+    assert body["priority"] == 2
+    assert body["status"] == "delivered"
+    mid = body["id"]
+
+    # sender can read it
+    assert client.get(f"/messages/{mid}", headers=h).status_code == 200
+    # receiver can read it
+    hb = auth(client, "bob")
+    assert client.get(f"/messages/{mid}", headers=hb).status_code == 200
+    # a third party cannot
+    hc = auth(client, "carol")
+    assert client.get(f"/messages/{mid}", headers=hc).status_code == 403
+    # admin can
+    assert client.get(f"/messages/{mid}",
+                      headers=auth(client, "admin")).status_code == 200
+    # 404 for unknown id
+    assert client.get("/messages/nope", headers=h).status_code == 404
+
+
+def test_receive_flow(client):
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    client.post("/messages", headers=ha,
+                json={"receiver_id": "bob", "content": "one"})
+    client.post("/messages", headers=ha,
+                json={"receiver_id": "bob", "content": {"n": 2}})
+    r = client.post("/agents/receive?timeout=0", headers=hb)
+    assert r.status_code == 200
+    msgs = r.json()
+    assert [m["content"] for m in msgs] == ["one", {"n": 2}]
+    assert all(m["status"] == "read" for m in msgs)
+    # drained
+    assert client.post("/agents/receive?timeout=0", headers=hb).json() == []
+
+
+def test_broadcast_endpoint(client):
+    for a in ["a", "b", "c"]:
+        client.post("/agents/register", headers=auth(client, a),
+                    json={"agent_id": a})
+    r = client.post("/messages/broadcast", headers=auth(client, "a"),
+                    json={"content": "all hands", "exclude_agents": ["c"]})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["status"] == "broadcast"
+    mid = body["message_id"]
+    got_b = client.post("/agents/receive?timeout=0", headers=auth(client, "b")).json()
+    assert [m["id"] for m in got_b] == [mid]
+    assert client.post("/agents/receive?timeout=0",
+                       headers=auth(client, "c")).json() == []
+
+
+def test_query_scope_enforcement(client):
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    client.post("/messages", headers=ha,
+                json={"receiver_id": "bob", "content": "x"})
+    client.post("/messages", headers=hb,
+                json={"receiver_id": "alice", "content": "y"})
+    # non-admin with no filter defaults to own sent messages
+    r = client.get("/messages", headers=ha)
+    assert r.status_code == 200
+    assert [m["content"] for m in r.json()] == ["x"]
+    # non-admin cannot query someone else's traffic
+    r = client.get("/messages?sender_id=bob", headers=ha)
+    assert r.status_code == 403
+    # receiver scope allowed
+    r = client.get("/messages?receiver_id=alice", headers=ha)
+    assert [m["content"] for m in r.json()] == ["y"]
+    # admin sees all
+    r = client.get("/messages", headers=auth(client, "admin"))
+    assert len(r.json()) == 2
+    # filters pass through
+    r = client.get("/messages?status=delivered&limit=1",
+                   headers=auth(client, "admin"))
+    assert len(r.json()) == 1
+
+
+def test_agent_messages_endpoint(client):
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    for i in range(3):
+        client.post("/messages", headers=ha,
+                    json={"receiver_id": "bob", "content": f"m{i}"})
+    r = client.get("/agents/bob/messages", headers=hb)
+    assert [m["content"] for m in r.json()] == ["m2", "m1", "m0"]
+    r = client.get("/agents/bob/messages?limit=1&skip=1", headers=hb)
+    assert [m["content"] for m in r.json()] == ["m1"]
+    assert client.get("/agents/bob/messages", headers=ha).status_code == 403
+    assert client.get("/agents/bob/messages",
+                      headers=auth(client, "admin")).status_code == 200
+
+
+def test_update_status_permissions(client):
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    mid = client.post("/messages", headers=ha,
+                      json={"receiver_id": "bob", "content": "x"}).json()["id"]
+    # sender (not receiver) cannot update
+    r = client.put(f"/messages/{mid}/status?status=read", headers=ha)
+    assert r.status_code == 403
+    r = client.put(f"/messages/{mid}/status?status=processed", headers=hb)
+    assert r.status_code == 200
+    assert client.get(f"/messages/{mid}", headers=hb).json()["status"] == "processed"
+    assert client.put("/messages/nope/status?status=read",
+                      headers=hb).status_code == 404
+
+
+def test_group_flow(client):
+    h = auth(client, "lead")
+    r = client.post("/groups", headers=h,
+                    json={"group_name": "team",
+                          "agent_ids": ["lead", "m1", "m2"]})
+    assert r.status_code == 201
+    r = client.post("/groups/message", headers=h,
+                    json={"group_name": "team", "content": "standup"})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["status"] == "sent"
+    assert len(body["message_ids"]) == 2
+    r = client.post("/groups/message", headers=h,
+                    json={"group_name": "ghost", "content": "x"})
+    assert r.status_code == 404
+
+
+def test_health_no_auth(client):
+    r = client.get("/health")
+    assert r.status_code == 200
+    body = r.json()
+    assert body["status"] == "healthy"
+    assert body["engine"] == "CpuEngine"
+    assert body["engine_connected"] is True
+
+
+def test_stats_admin_only(client):
+    assert client.get("/stats", headers=auth(client, "pleb")).status_code == 403
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "x"})
+    r = client.get("/stats", headers=auth(client, "admin"))
+    assert r.status_code == 200
+    assert r.json()["total_messages"] == 1
+
+
+def test_admin_save_flush_resend_scale(client):
+    ha = auth(client, "admin")
+    hx = auth(client, "pleb")
+    for path in ["/admin/save", "/admin/flush", "/admin/resend_failed",
+                 "/admin/scale_partitions"]:
+        assert client.post(path, headers=hx).status_code == 403
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "x"})
+    r = client.post("/admin/save", headers=ha)
+    assert r.status_code == 200
+    assert Path(r.json()["path"]).exists()
+    r = client.post("/admin/flush?older_than=0.0", headers=ha)
+    assert r.status_code == 200
+    assert r.json()["messages_flushed"] == 1
+    r = client.post("/admin/resend_failed", headers=ha)
+    assert r.json() == {"status": "resent", "message_ids": []}
+    r = client.post("/admin/scale_partitions", headers=ha)
+    assert r.status_code == 200
+    assert "current_partitions" in r.json()
+
+
+def test_admin_load_endpoint(client):
+    ha = auth(client, "admin")
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "persist me"})
+    path = client.post("/admin/save", headers=ha).json()["path"]
+    r = client.post(f"/admin/load?path={path}", headers=ha)
+    assert r.status_code == 200
+    r = client.post("/admin/load?path=/nonexistent.json", headers=ha)
+    assert r.status_code == 404
+
+
+def test_llm_routes(client):
+    ha = auth(client, "admin")
+    r = client.post("/llm/dispatch", headers=ha)
+    assert r.status_code == 503  # no backends yet
+    for b in ["gpu0", "gpu1"]:
+        assert client.post(f"/llm/backends/{b}", headers=ha).status_code == 200
+    client.db.set_llm_load_balancing(True)
+    picks = [client.post("/llm/dispatch", headers=ha).json()["backend_id"]
+             for _ in range(4)]
+    assert sorted(set(picks)) == ["gpu0", "gpu1"]
+    r = client.post("/llm/complete/gpu0", headers=ha)
+    assert r.status_code == 200
+
+
+def test_load_and_unread_routes(client):
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    client.post("/messages", headers=ha,
+                json={"receiver_id": "bob", "content": "x"})
+    r = client.get("/agents/bob/unread_count", headers=hb)
+    assert r.json()["unread_count"] == 1
+    r = client.get("/agents/bob/load", headers=hb)
+    assert r.json()["inbox_size"] == 1
+    assert client.get("/agents/bob/load", headers=ha).status_code == 403
+
+
+def test_rate_limiter():
+    import swarmdb_amd.api.app as appmod
+
+    rl = appmod.RateLimiter(limit_per_minute=3)
+    assert all(rl.allow("1.2.3.4") for _ in range(3))
+    assert not rl.allow("1.2.3.4")
+    assert rl.allow("5.6.7.8")  # other IPs unaffected
