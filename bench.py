@@ -1,0 +1,248 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — messages/sec (whole node) + p50 send->receive
+latency, 1024 agents per GPU (BASELINE.json headline metric).
+
+Each timed step is one full delivery tick through the GPU-resident queue:
+
+  build batch -> [N>1: RCCL all-to-all routing over xGMI] -> pinned H2D
+  staging -> enqueue kernel (slot write + inbox append + DELIVERED ack)
+  -> broadcast fan-out kernel -> dequeue kernel (visibility filter +
+  LDS sort + READ) -> payload gather kernel -> D2H -> host holds bytes.
+
+A message's send->receive latency is bounded by the tick duration, so the
+reported p50 latency is the median timed step duration.
+
+Usage (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches via torch.distributed.run; ranks read
+RANK/LOCAL_RANK/WORLD_SIZE from the env. Scaling is weak: 1024 agents and
+the same message load per GPU.
+
+No GPU present -> falls back to the CPU engine (config-1 loopback) so the
+script stays runnable in CPU CI; the driver's numbers come from MI355X.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import numpy as np
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+from swarmdb_amd.core.config import QueueConfig  # noqa: E402
+from swarmdb_amd.runtime.engine import (  # noqa: E402
+    BROADCAST,
+    NO_BITMAP,
+    REC_DTYPE,
+    VIS_ALL,
+)
+
+
+def build_batches(rng, n_batches, batch, agents_global, rank, world,
+                  payload_bytes, bcast_frac=0.0):
+    """Pre-build rotating synthetic batches (random agent ids, 1 KB chat
+    messages — BASELINE 'synthetic messages, random agent IDs')."""
+    stride = (payload_bytes + 15) // 16 * 16
+    local_agents = np.arange(rank, agents_global, world, dtype=np.uint32)
+    batches = []
+    for _ in range(n_batches):
+        recs = np.zeros(batch, dtype=REC_DTYPE)
+        # senders are local agents; receivers uniform over the global space
+        recs["sender"] = rng.choice(local_agents, batch)
+        recv = rng.integers(0, agents_global, batch).astype(np.uint32)
+        if bcast_frac > 0:
+            recv[rng.random(batch) < bcast_frac] = BROADCAST
+        recs["receiver"] = recv
+        recs["type"] = 0  # chat
+        recs["priority"] = rng.integers(0, 4, batch)
+        recs["timestamp"] = time.time()
+        recs["vis_mode"] = VIS_ALL
+        recs["bitmap"] = NO_BITMAP
+        recs["payload_len"] = payload_bytes
+        recs["content_len"] = payload_bytes
+        recs["payload_off"] = np.arange(batch, dtype=np.uint64) * stride
+        payload = rng.integers(32, 127, batch * stride, dtype=np.uint8).tobytes()
+        batches.append((recs, payload))
+    return batches, local_agents
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--agents", type=int, default=1024,
+                    help="agents per GPU (weak scaling)")
+    ap.add_argument("--batch", type=int, default=16384,
+                    help="messages per step per GPU")
+    ap.add_argument("--payload", type=int, default=1024,
+                    help="payload bytes (1 KB chat messages)")
+    ap.add_argument("--priority", action="store_true",
+                    help="priority-ordered dequeue (config 3 kernel)")
+    ap.add_argument("--no-gather", action="store_true",
+                    help="skip payload D2H gather (delivery stays device-side)")
+    args = ap.parse_args()
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    have_gpu = torch.cuda.is_available()
+
+    dist_on = world > 1
+    if dist_on:
+        import torch.distributed as dist
+
+        backend = "nccl" if have_gpu else "gloo"
+        if have_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    agents_global = args.agents * world
+    max_agents = ((agents_global + 63) // 64) * 64
+    rng = np.random.default_rng(1234 + rank)
+
+    cfg = QueueConfig(
+        use_gpu=have_gpu,
+        max_agents=max_agents,
+        num_slots=max(1 << 20, 4 * args.batch * world),
+        slot_bytes=max(2048, ((args.payload + 63) // 16) * 16 + 64),
+        inbox_capacity=1 << 16,
+        staging_batch=max(16384, args.batch * (2 if dist_on else 1)),
+        device_index=local_rank,
+        auto_save=False,
+        world_size=world,
+        rank=rank,
+    )
+    if have_gpu:
+        from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+        engine = GpuEngine(cfg)
+    else:
+        from swarmdb_amd.runtime.cpu_engine import CpuEngine
+
+        engine = CpuEngine(cfg)
+
+    batches, local_agents = build_batches(
+        rng, min(8, max(2, args.steps)), args.batch, agents_global, rank,
+        world, args.payload,
+    )
+    for a in local_agents:
+        engine.register_agent(int(a))
+
+    router = None
+    if dist_on:
+        from swarmdb_amd.parallel.router import CrossGpuRouter
+
+        device = torch.device("cuda", local_rank) if have_gpu else torch.device("cpu")
+        router = CrossGpuRouter(device)
+
+    recv_K = max(64, 4 * args.batch * world // max(1, len(local_agents)))
+    sent_total = 0
+    recv_total = 0
+
+    def barrier_sync():
+        if dist_on:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    def step(i: int) -> int:
+        nonlocal sent_total, recv_total
+        recs, payload = batches[i % len(batches)]
+        if router is not None:
+            recs, payload = router.route(recs, payload)
+        engine.enqueue_batch(recs, payload)
+        sent_local = len(recs)
+        counts, seqs = engine.receive_many(
+            local_agents, recv_K, priority_order=args.priority
+        )
+        ndel = int(counts.sum())
+        if not args.no_gather and ndel:
+            engine.fetch(seqs)  # payload gather + D2H: bytes land on host
+        sent_total += sent_local
+        recv_total += ndel
+        return ndel
+
+    for i in range(args.warmup):
+        step(i)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    step_times = []
+    for i in range(args.steps):
+        s = time.perf_counter()
+        step(args.warmup + i)
+        step_times.append(time.perf_counter() - s)
+    barrier_sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if dist_on:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        tot = torch.tensor([float(recv_total)], dtype=torch.float64,
+                           device="cuda" if have_gpu else "cpu")
+        dist.all_reduce(tot, op=dist.ReduceOp.SUM)
+        recv_all = int(tot.item())
+    else:
+        recv_all = recv_total
+
+    msgs_per_s = recv_all / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+    p50_ms = float(np.median(step_times) * 1000.0)
+
+    if rank == 0:
+        out = {
+            "metric": "messages/sec (whole node), 1024 agents/GPU, "
+                      "1 KB synthetic chat messages",
+            "value": round(msgs_per_s, 1),
+            "unit": "msg/s",
+            "n_gpus": world if have_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "uint8",
+            "data": "synthetic",
+            "p50_latency_ms": round(p50_ms, 3),
+            "config": {
+                "model": "gpu-mpmc-ring-queue",
+                "global_batch": args.batch * world,
+                "seq_len": args.payload,
+                "parallelism": f"agent-sharded all-to-all x{world}"
+                if world > 1
+                else "single-gpu",
+                "agents": agents_global,
+                "engine": type(engine).__name__,
+                "priority_dequeue": bool(args.priority),
+                "payload_gather_d2h": not args.no_gather,
+            },
+        }
+        print(json.dumps(out))
+
+    engine.close()
+    if dist_on:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,1096 @@
+// swarmq_module.hip — MI355X-native message-queue engine.
+//
+// HBM-resident MPMC slot ring + per-agent inbox rings with hand-written
+// CDNA4 HIP kernels (gfx950): wavefront-cooperative enqueue (64 lanes x
+// 16 B = 1 KiB per copy round), broadcast fan-out, LDS-bitonic priority
+// dequeue, filter/search scans, and a wavefront-shuffle least-loaded
+// reduction for LLM backend dispatch.
+//
+// This is the native replacement for the reference's librdkafka/Kafka
+// tier (reference "swarmdb/ main.py":192-207, 334-345, 466-484, 553-588;
+// SURVEY.md §2.4). All kernels run on one HIP stream per queue, so
+// cross-kernel visibility is stream-ordered — no inter-workgroup
+// release/acquire protocol is needed inside a launch (each workgroup
+// owns disjoint state: a message, or an agent).
+//
+// Build: hipcc --offload-arch=gfx950 (see setup.py) — no CUDA paths, no
+// hipify, no Triton.
+
+#include <hip/hip_runtime.h>
+
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <algorithm>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "swarmq_common.h"
+
+namespace py = pybind11;
+using namespace swarmq;
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string("HIP error at " __FILE__ ":") +     \
+                               std::to_string(__LINE__) + ": " +               \
+                               hipGetErrorString(_e));                         \
+    }                                                                          \
+  } while (0)
+
+using ull = unsigned long long;
+
+// ---------------------------------------------------------------------------
+// kernels
+// ---------------------------------------------------------------------------
+
+// Enqueue a staged batch: one 64-lane wavefront per message. The wave
+// cooperatively copies the payload in 16-B chunks (64 lanes x 16 B = 1 KiB
+// per round — a 1 KiB chat message is ONE vector round); lane 0 writes the
+// header, status word (the DELIVERED ack, replacing the Kafka delivery
+// callback, reference "swarmdb/ main.py":374-391), counters, and the
+// point-to-point inbox append. Broadcast messages are queued into
+// bcast_list for k_fanout.
+__global__ void k_enqueue(const Rec *__restrict__ stage,
+                          const u8 *__restrict__ stage_pay, int n,
+                          u64 base_seq, Rec *__restrict__ hdr,
+                          u32 *__restrict__ status, u8 *__restrict__ payload,
+                          u64 *__restrict__ inbox, ull *__restrict__ inbox_wpos,
+                          ull *__restrict__ by_type,
+                          ull *__restrict__ by_status, ull *__restrict__ sent,
+                          u64 *__restrict__ bcast_list,
+                          u32 *__restrict__ bcast_count, QueueGeom g) {
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (wave >= n)
+    return;
+  const Rec r = stage[wave];
+  const u64 seq = base_seq + (u64)wave;
+  const u32 slot = (u32)(seq % g.num_slots);
+
+  // payload copy, 16 B per lane per round (staging offsets are 16-B
+  // aligned; slots are slot_bytes-strided so destination is aligned too)
+  const uint4 *src =
+      reinterpret_cast<const uint4 *>(stage_pay + r.payload_off);
+  uint4 *dst =
+      reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
+  const u32 nchunk = (r.payload_len + 15u) >> 4;
+  for (u32 c = lane; c < nchunk; c += 64)
+    dst[c] = src[c];
+
+  if (lane == 0) {
+    Rec h = r;
+    h.payload_off = (u64)slot * g.slot_bytes;
+    hdr[slot] = h;
+    status[slot] = ST_DELIVERED;
+    atomicAdd(&by_type[r.type], 1ull);
+    atomicAdd(&by_status[ST_DELIVERED], 1ull);
+    atomicAdd(&sent[r.sender], 1ull);
+    if (r.receiver == BROADCAST) {
+      u32 bi = atomicAdd(bcast_count, 1u);
+      bcast_list[bi] = seq;
+    } else {
+      ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
+      inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
+          seq;
+    }
+  }
+}
+
+// Broadcast fan-out: one thread per agent appends the batch's broadcast
+// seqs to its own inbox ring — no atomics (each thread owns its agent,
+// and kernels are stream-serialized). Replaces the reference's Python
+// loop over all inboxes (reference "swarmdb/ main.py":457-463).
+__global__ void k_fanout(const u64 *__restrict__ bcast_list,
+                         const u32 *__restrict__ bcast_count,
+                         const u32 *__restrict__ active,
+                         u64 *__restrict__ inbox, ull *__restrict__ inbox_wpos,
+                         QueueGeom g) {
+  const u32 a = blockIdx.x * blockDim.x + threadIdx.x;
+  if (a >= g.max_agents || !active[a])
+    return;
+  const u32 nb = *bcast_count;
+  if (nb == 0)
+    return;
+  const ull base = inbox_wpos[a];
+  u64 *ib = inbox + (u64)a * g.inbox_capacity;
+  for (u32 i = 0; i < nb; ++i)
+    ib[(base + i) % g.inbox_capacity] = bcast_list[i];
+  inbox_wpos[a] = base + nb;
+}
+
+// Dequeue: one 256-thread workgroup per polling agent. Drains the
+// agent's carry buffer + fresh inbox window into LDS, applies the
+// visibility filter on-device (reference's client-side filter at
+// "swarmdb/ main.py":579-585 moved into the kernel — each message is
+// touched O(recipients) times instead of O(agents)), bitonic-sorts the
+// window (by seq for FIFO, or (3-priority)<<48|seq for priority mode —
+// the priority-sort kernel of BASELINE config 3), marks the delivered
+// prefix READ, and carries leftovers.
+__global__ void __launch_bounds__(256)
+    k_receive(const u32 *__restrict__ agents, int n_agents, int max_per_agent,
+              int priority_mode, u64 evict_base, const Rec *__restrict__ hdr,
+              u32 *__restrict__ status, const u64 *__restrict__ inbox,
+              ull *__restrict__ inbox_wpos, ull *__restrict__ inbox_rpos,
+              u64 *__restrict__ carry, u32 *__restrict__ carry_n,
+              const u64 *__restrict__ bitmaps, u64 *__restrict__ out_seqs,
+              u32 *__restrict__ out_counts, ull *__restrict__ by_status,
+              ull *__restrict__ received, QueueGeom g) {
+  __shared__ u64 keys[RECV_WINDOW]; // 32 KiB of 160 KiB LDS
+  __shared__ u32 sh_valid;
+
+  const int b = blockIdx.x;
+  if (b >= n_agents)
+    return;
+  const u32 a = agents[b];
+  const ull r = inbox_rpos[a];
+  const ull w = inbox_wpos[a];
+  const u32 nc = carry_n[a];
+  const u32 room = g.recv_window - nc;
+  const ull avail = w - r;
+  const u32 fresh = (u32)(avail < (ull)room ? avail : (ull)room);
+  const u32 total = nc + fresh;
+
+  const u64 *ib = inbox + (u64)a * g.inbox_capacity;
+  const u64 *cb = carry + (u64)a * g.recv_window;
+
+  for (u32 i = threadIdx.x; i < total; i += blockDim.x) {
+    const u64 seq =
+        (i < nc) ? cb[i] : ib[(r + (i - nc)) % g.inbox_capacity];
+    u64 key = KEY_INVALID;
+    if (seq >= evict_base) {
+      const u32 slot = (u32)(seq % g.num_slots);
+      if (status[slot] != ST_DELETED) {
+        const Rec h = hdr[slot];
+        bool vis = true;
+        if (h.vis_mode == VIS_BITMAP && h.bitmap != NO_BITMAP) {
+          const u64 wbits =
+              bitmaps[(u64)h.bitmap * g.bitmap_words + (a >> 6)];
+          vis = (wbits >> (a & 63)) & 1ull;
+        }
+        if (vis)
+          key = priority_mode ? prio_key(h.priority, seq) : seq;
+      }
+    }
+    keys[i] = key;
+  }
+  u32 npad = 1;
+  while (npad < total)
+    npad <<= 1;
+  if (total == 0)
+    npad = 0;
+  for (u32 i = threadIdx.x; i < npad; i += blockDim.x)
+    if (i >= total)
+      keys[i] = KEY_INVALID;
+  __syncthreads();
+
+  // bitonic sort ascending (invalid keys sink to the end)
+  for (u32 k = 2; k <= npad; k <<= 1) {
+    for (u32 j = k >> 1; j > 0; j >>= 1) {
+      for (u32 i = threadIdx.x; i < npad; i += blockDim.x) {
+        const u32 ixj = i ^ j;
+        if (ixj > i) {
+          const u64 x = keys[i], y = keys[ixj];
+          const bool up = ((i & k) == 0);
+          if ((x > y) == up) {
+            keys[i] = y;
+            keys[ixj] = x;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  if (threadIdx.x == 0) {
+    u32 lo = 0, hi = npad;
+    while (lo < hi) { // first INVALID = count of deliverable keys
+      const u32 mid = (lo + hi) >> 1;
+      if (keys[mid] == KEY_INVALID)
+        hi = mid;
+      else
+        lo = mid + 1;
+    }
+    sh_valid = lo;
+  }
+  __syncthreads();
+  const u32 valid = sh_valid;
+  const u32 take = valid < (u32)max_per_agent ? valid : (u32)max_per_agent;
+
+  for (u32 i = threadIdx.x; i < take; i += blockDim.x) {
+    const u64 seq =
+        priority_mode ? (keys[i] & 0xFFFFFFFFFFFFull) : keys[i];
+    out_seqs[(u64)b * (u32)max_per_agent + i] = seq;
+    const u32 slot = (u32)(seq % g.num_slots);
+    // DELIVERED -> READ transition; CAS because two agents may read the
+    // same broadcast slot concurrently (status is global, as in the
+    // reference)
+    const u32 old = atomicCAS(&status[slot], ST_DELIVERED, ST_READ);
+    if (old == ST_DELIVERED) {
+      atomicAdd(&by_status[ST_READ], 1ull);
+      atomicAdd(&by_status[ST_DELIVERED], (ull)(-1ll));
+    }
+  }
+
+  const u32 rest = valid - take;
+  u64 *cbw = carry + (u64)a * g.recv_window;
+  for (u32 i = threadIdx.x; i < rest; i += blockDim.x) {
+    const u64 key = keys[take + i];
+    cbw[i] = priority_mode ? (key & 0xFFFFFFFFFFFFull) : key;
+  }
+  if (threadIdx.x == 0) {
+    carry_n[a] = rest;
+    inbox_rpos[a] = r + fresh;
+    out_counts[b] = take;
+    if (take)
+      atomicAdd(&received[a], (ull)take);
+  }
+}
+
+// Gather message contents for the host (receive payload delivery, fetch,
+// history spill): one wave per message, dense slot_bytes-strided output.
+__global__ void k_gather(const u64 *__restrict__ seqs, int n,
+                         const Rec *__restrict__ hdr,
+                         const u32 *__restrict__ status,
+                         const u8 *__restrict__ payload,
+                         Rec *__restrict__ out_hdr,
+                         u32 *__restrict__ out_status,
+                         u8 *__restrict__ out_pay, u64 evict_base,
+                         QueueGeom g) {
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (wave >= n)
+    return;
+  const u64 seq = seqs[wave];
+  const u32 slot = (u32)(seq % g.num_slots);
+  if (seq < evict_base) {
+    if (lane == 0) {
+      Rec z{};
+      out_hdr[wave] = z;
+      out_status[wave] = ST_DELETED;
+    }
+    return;
+  }
+  const Rec h = hdr[slot];
+  const uint4 *src = reinterpret_cast<const uint4 *>(payload + h.payload_off);
+  uint4 *dst =
+      reinterpret_cast<uint4 *>(out_pay + (u64)wave * g.slot_bytes);
+  const u32 nchunk = (h.payload_len + 15u) >> 4;
+  for (u32 c = lane; c < nchunk; c += 64)
+    dst[c] = src[c];
+  if (lane == 0) {
+    out_hdr[wave] = h;
+    out_status[wave] = status[slot];
+  }
+}
+
+// Filtered scan over a seq range (query_messages engine, reference
+// "swarmdb/ main.py":671-740): one thread per message, matches appended
+// with a global atomic; the host sorts newest-first.
+__global__ void k_filter(u64 lo, u64 hi, int f_sender, int f_receiver,
+                         int f_type, int f_status, double after, double before,
+                         u32 pred_mask, const Rec *__restrict__ hdr,
+                         const u32 *__restrict__ status,
+                         u64 *__restrict__ out, u32 *__restrict__ out_count,
+                         u32 cap, QueueGeom g) {
+  const u64 stride = (u64)gridDim.x * blockDim.x;
+  for (u64 seq = lo + blockIdx.x * blockDim.x + threadIdx.x; seq < hi;
+       seq += stride) {
+    const u32 slot = (u32)(seq % g.num_slots);
+    const u32 st = status[slot];
+    if (st == ST_DELETED)
+      continue;
+    const Rec h = hdr[slot];
+    if ((pred_mask & 1u) && h.sender != (u32)f_sender)
+      continue;
+    if ((pred_mask & 2u) && h.receiver != (u32)f_receiver)
+      continue;
+    if ((pred_mask & 4u) && h.type != (u8)f_type)
+      continue;
+    if ((pred_mask & 8u) && st != (u32)f_status)
+      continue;
+    if ((pred_mask & 16u) && !(h.timestamp > after)) // exclusive bounds
+      continue;
+    if ((pred_mask & 32u) && !(h.timestamp < before))
+      continue;
+    const u32 i = atomicAdd(out_count, 1u);
+    if (i < cap)
+      out[i] = seq;
+  }
+}
+
+// Substring search over message content (search_messages engine,
+// reference "swarmdb/ main.py":742-781): one wave per message; lanes
+// stride over candidate start positions; ASCII case-folding optional.
+// The content window is payload[0:content_len] — metadata/ids never
+// match.
+__global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
+                         int nlen, int fold, const Rec *__restrict__ hdr,
+                         const u32 *__restrict__ status,
+                         const u8 *__restrict__ payload,
+                         u64 *__restrict__ out, u32 *__restrict__ out_count,
+                         u32 cap, QueueGeom g) {
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  const int waves = gridDim.x * (blockDim.x >> 6);
+  for (u64 seq = lo + wave; seq < hi; seq += waves) {
+    const u32 slot = (u32)(seq % g.num_slots);
+    if (status[slot] == ST_DELETED)
+      continue;
+    const Rec h = hdr[slot];
+    if ((int)h.content_len < nlen)
+      continue;
+    const u8 *text = payload + h.payload_off;
+    bool found = false;
+    const int nstart = (int)h.content_len - nlen + 1;
+    for (int p = lane; p < nstart && !found; p += 64) {
+      bool m = true;
+      for (int q = 0; q < nlen; ++q) {
+        u8 c = text[p + q], d = needle[q];
+        if (fold) {
+          c |= (c >= 'A' && c <= 'Z') ? 0x20 : 0;
+          d |= (d >= 'A' && d <= 'Z') ? 0x20 : 0;
+        }
+        if (c != d) {
+          m = false;
+          break;
+        }
+      }
+      if (m)
+        found = true;
+    }
+    if (__any(found) && lane == 0) {
+      const u32 i = atomicAdd(out_count, 1u);
+      if (i < cap)
+        out[i] = seq;
+    }
+  }
+}
+
+// Unread count: one workgroup per agent scans its retained inbox window
+// for entries whose global status is DELIVERED (reference "swarmdb/
+// main.py":1026-1047).
+__global__ void k_unread(const u32 *__restrict__ agents, int n_agents,
+                         u64 evict_base, const u32 *__restrict__ status,
+                         const u64 *__restrict__ inbox,
+                         const ull *__restrict__ inbox_wpos,
+                         u32 *__restrict__ out, QueueGeom g) {
+  __shared__ u32 cnt;
+  const int b = blockIdx.x;
+  if (b >= n_agents)
+    return;
+  if (threadIdx.x == 0)
+    cnt = 0;
+  __syncthreads();
+  const u32 a = agents[b];
+  const ull w = inbox_wpos[a];
+  const ull start = w > g.inbox_capacity ? w - g.inbox_capacity : 0;
+  const u64 *ib = inbox + (u64)a * g.inbox_capacity;
+  u32 local = 0;
+  for (ull i = start + threadIdx.x; i < w; i += blockDim.x) {
+    const u64 seq = ib[i % g.inbox_capacity];
+    if (seq < evict_base)
+      continue;
+    if (status[seq % g.num_slots] == ST_DELIVERED)
+      ++local;
+  }
+  atomicAdd(&cnt, local);
+  __syncthreads();
+  if (threadIdx.x == 0)
+    out[b] = cnt;
+}
+
+// Status mutation with counter upkeep (mark processed / admin status
+// updates / delete tombstones).
+__global__ void k_set_status(u64 seq, u32 new_status, u32 *__restrict__ status,
+                             ull *__restrict__ by_status, QueueGeom g) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    const u32 slot = (u32)(seq % g.num_slots);
+    const u32 old = atomicExch(&status[slot], new_status);
+    if (old != new_status) {
+      if (old < N_STATUS)
+        atomicAdd(&by_status[old], (ull)(-1ll));
+      if (new_status < N_STATUS)
+        atomicAdd(&by_status[new_status], 1ull);
+    }
+  }
+}
+
+// Least-loaded dispatch (BASELINE config 5): one wavefront holds all
+// backend loads in registers; each request is an exact sequential
+// argmin via a 64-lane shuffle min-reduce, then the winner's load is
+// bumped — the CDNA4 reduction-kernel mechanism the reference never
+// implemented (SURVEY.md §2.2 "LLM load balancing").
+__global__ void k_lb_batch(int requests, int n_backends,
+                           ull *__restrict__ loads,
+                           u32 *__restrict__ choices) {
+  const int lane = threadIdx.x;
+  if (lane >= 64)
+    return;
+  long long my = (lane < n_backends) ? (long long)loads[lane]
+                                     : 0x7FFFFFFFFFFFFFFFll;
+  for (int r = 0; r < requests; ++r) {
+    long long v = my;
+    int li = lane;
+    for (int o = 32; o > 0; o >>= 1) {
+      const long long ov = __shfl_down(v, o, 64);
+      const int ol = __shfl_down(li, o, 64);
+      if (ov < v || (ov == v && ol < li)) {
+        v = ov;
+        li = ol;
+      }
+    }
+    li = __shfl(li, 0, 64);
+    if (lane == li)
+      ++my;
+    if (lane == 0 && choices != nullptr)
+      choices[r] = (u32)li;
+  }
+  if (lane < n_backends)
+    loads[lane] = (ull)my;
+}
+
+__global__ void k_add_load(int idx, long long delta, ull *__restrict__ loads) {
+  if (threadIdx.x == 0 && blockIdx.x == 0)
+    atomicAdd(&loads[idx], (ull)delta);
+}
+
+// ---------------------------------------------------------------------------
+// DeviceQueue — host runtime (the librdkafka-equivalent native layer)
+// ---------------------------------------------------------------------------
+
+class DeviceQueue {
+public:
+  DeviceQueue(u32 num_slots, u32 slot_bytes, u32 max_agents,
+              u32 inbox_capacity, u32 num_bitmaps, u32 num_backends,
+              u32 staging_batch, int device)
+      : device_(device), staging_batch_(staging_batch) {
+    if (slot_bytes % 16 != 0)
+      throw std::invalid_argument("slot_bytes must be a multiple of 16");
+    if (max_agents % 64 != 0)
+      throw std::invalid_argument("max_agents must be a multiple of 64");
+    g_.num_slots = num_slots;
+    g_.slot_bytes = slot_bytes;
+    g_.max_agents = max_agents;
+    g_.inbox_capacity = inbox_capacity;
+    g_.num_bitmaps = num_bitmaps;
+    g_.bitmap_words = max_agents / 64;
+    g_.num_backends = num_backends;
+    g_.recv_window = RECV_WINDOW;
+
+    HIP_CHECK(hipSetDevice(device_));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_, hipEventDisableTiming));
+
+    // device state
+    HIP_CHECK(hipMalloc(&d_hdr_, (size_t)num_slots * sizeof(Rec)));
+    HIP_CHECK(hipMalloc(&d_status_, (size_t)num_slots * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_payload_, (size_t)num_slots * slot_bytes));
+    HIP_CHECK(
+        hipMalloc(&d_inbox_, (size_t)max_agents * inbox_capacity * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_wpos_, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_rpos_, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(
+        hipMalloc(&d_carry_, (size_t)max_agents * RECV_WINDOW * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_carry_n_, (size_t)max_agents * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_active_, (size_t)max_agents * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_bitmaps_,
+                        (size_t)num_bitmaps * g_.bitmap_words * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_by_type_, N_TYPES * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_by_status_, N_STATUS * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_sent_, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_received_, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_bcast_, (size_t)staging_batch * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_bcast_count_, sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_backend_loads_, (size_t)num_backends * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_choices_, (size_t)staging_batch * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_match_, (size_t)staging_batch * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_match_count_, sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_needle_, 256));
+    HIP_CHECK(hipMalloc(&d_agents_, (size_t)max_agents * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_unread_, (size_t)max_agents * sizeof(u32)));
+
+    out_pool_ = (size_t)4 << 20; // receive output pool: 4M entries, 32 MB
+    HIP_CHECK(hipMalloc(&d_out_seqs_, out_pool_ * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_out_counts_, (size_t)max_agents * sizeof(u32)));
+
+    // staging (device side of the H2D batch)
+    HIP_CHECK(hipMalloc(&d_stage_recs_, (size_t)staging_batch * sizeof(Rec)));
+    stage_pay_bytes_ = (size_t)staging_batch * 1024; // grows on demand
+    HIP_CHECK(hipMalloc(&d_stage_pay_, stage_pay_bytes_));
+    HIP_CHECK(hipMalloc(&d_seqs_in_, (size_t)staging_batch * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_fetch_hdr_, (size_t)staging_batch * sizeof(Rec)));
+    HIP_CHECK(hipMalloc(&d_fetch_status_, (size_t)staging_batch * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_fetch_pay_, (size_t)staging_batch * slot_bytes));
+
+    // pinned host staging
+    HIP_CHECK(hipHostMalloc(&h_recs_, (size_t)staging_batch * sizeof(Rec)));
+    HIP_CHECK(hipHostMalloc(&h_pay_, stage_pay_bytes_));
+    HIP_CHECK(hipHostMalloc(&h_out_seqs_, out_pool_ * sizeof(u64)));
+    HIP_CHECK(hipHostMalloc(&h_out_counts_, (size_t)max_agents * sizeof(u32)));
+    HIP_CHECK(hipHostMalloc(&h_fetch_hdr_, (size_t)staging_batch * sizeof(Rec)));
+    HIP_CHECK(
+        hipHostMalloc(&h_fetch_status_, (size_t)staging_batch * sizeof(u32)));
+    HIP_CHECK(hipHostMalloc(&h_fetch_pay_, (size_t)staging_batch * slot_bytes));
+    HIP_CHECK(hipHostMalloc(&h_choices_, (size_t)staging_batch * sizeof(u32)));
+
+    // zero all mutable state
+    HIP_CHECK(hipMemset(d_status_, 0, (size_t)num_slots * sizeof(u32)));
+    HIP_CHECK(hipMemset(d_wpos_, 0, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_rpos_, 0, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_carry_n_, 0, (size_t)max_agents * sizeof(u32)));
+    HIP_CHECK(hipMemset(d_active_, 0, (size_t)max_agents * sizeof(u32)));
+    HIP_CHECK(hipMemset(d_by_type_, 0, N_TYPES * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_by_status_, 0, N_STATUS * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_sent_, 0, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_received_, 0, (size_t)max_agents * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_backend_loads_, 0, (size_t)num_backends * sizeof(ull)));
+  }
+
+  ~DeviceQueue() { release(); }
+
+  void release() {
+    if (released_)
+      return;
+    released_ = true;
+    (void)hipStreamSynchronize(stream_);
+    (void)hipStreamSynchronize(copy_stream_);
+    for (void *p :
+         std::vector<void *>{d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
+                             d_rpos_, d_carry_, d_carry_n_, d_active_,
+                             d_bitmaps_, d_by_type_, d_by_status_, d_sent_,
+                             d_received_, d_bcast_, d_bcast_count_,
+                             d_backend_loads_, d_choices_, d_match_,
+                             d_match_count_, d_needle_, d_agents_, d_unread_,
+                             d_out_seqs_, d_out_counts_, d_stage_recs_,
+                             d_stage_pay_, d_seqs_in_, d_fetch_hdr_,
+                             d_fetch_status_, d_fetch_pay_})
+      (void)hipFree(p);
+    for (void *p : std::vector<void *>{h_recs_, h_pay_, h_out_seqs_,
+                                       h_out_counts_, h_fetch_hdr_,
+                                       h_fetch_status_, h_fetch_pay_,
+                                       h_choices_})
+      (void)hipHostFree(p);
+    (void)hipEventDestroy(ev_);
+    (void)hipStreamDestroy(stream_);
+    (void)hipStreamDestroy(copy_stream_);
+  }
+
+  // ---- registry ----
+
+  void register_agent(u32 idx) {
+    check_agent(idx);
+    const u32 one = 1;
+    HIP_CHECK(hipMemcpy(d_active_ + idx, &one, sizeof(u32),
+                        hipMemcpyHostToDevice));
+  }
+
+  void deregister_agent(u32 idx) {
+    check_agent(idx);
+    const u32 zero = 0;
+    HIP_CHECK(hipMemcpy(d_active_ + idx, &zero, sizeof(u32),
+                        hipMemcpyHostToDevice));
+  }
+
+  py::array_t<u32> active_agents() {
+    py::array_t<u32> out(g_.max_agents);
+    HIP_CHECK(hipMemcpy(out.mutable_data(), d_active_,
+                        g_.max_agents * sizeof(u32), hipMemcpyDeviceToHost));
+    return out;
+  }
+
+  // ---- send plane ----
+
+  // recs: n x 48 bytes (REC_DTYPE), payload_off 16-B aligned into `pay`.
+  // Returns base seq. The stream sync at the end is the DELIVERED ack.
+  u64 enqueue_batch(py::buffer recs, py::buffer pay, int n) {
+    py::buffer_info ri = recs.request(), pi = pay.request();
+    if ((size_t)ri.size * ri.itemsize < (size_t)n * sizeof(Rec))
+      throw std::invalid_argument("recs buffer too small");
+    if (n <= 0)
+      return count_;
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("batch exceeds staging_batch");
+    const size_t pay_bytes = (size_t)pi.size * pi.itemsize;
+    ensure_stage_pay(pay_bytes + 16);
+
+    const u64 base = count_;
+    {
+      py::gil_scoped_release nogil;
+      std::memcpy(h_recs_, ri.ptr, (size_t)n * sizeof(Rec));
+      if (pay_bytes)
+        std::memcpy(h_pay_, pi.ptr, pay_bytes);
+      HIP_CHECK(hipMemcpyAsync(d_stage_recs_, h_recs_, (size_t)n * sizeof(Rec),
+                               hipMemcpyHostToDevice, stream_));
+      if (pay_bytes)
+        HIP_CHECK(hipMemcpyAsync(d_stage_pay_, h_pay_, pay_bytes,
+                                 hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
+      const int waves_per_block = 4; // 256 threads
+      const int blocks = (n + waves_per_block - 1) / waves_per_block;
+      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
+                         d_stage_recs_, d_stage_pay_, n, base, d_hdr_,
+                         d_status_, d_payload_, d_inbox_, d_wpos_, d_by_type_,
+                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
+                         0, stream_, d_bcast_, d_bcast_count_, d_active_,
+                         d_inbox_, d_wpos_, g_);
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    count_ = base + (u64)n;
+    if (count_ > g_.num_slots)
+      evict_base_ = count_ - g_.num_slots;
+    return base;
+  }
+
+  // Async variant for graph-friendly benchmarking: no sync; caller must
+  // call sync() before reading results.
+  u64 enqueue_batch_async(py::buffer recs, py::buffer pay, int n) {
+    py::buffer_info ri = recs.request(), pi = pay.request();
+    if (n <= 0)
+      return count_;
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("batch exceeds staging_batch");
+    const size_t pay_bytes = (size_t)pi.size * pi.itemsize;
+    ensure_stage_pay(pay_bytes + 16);
+    const u64 base = count_;
+    {
+      py::gil_scoped_release nogil;
+      std::memcpy(h_recs_, ri.ptr, (size_t)n * sizeof(Rec));
+      if (pay_bytes)
+        std::memcpy(h_pay_, pi.ptr, pay_bytes);
+      HIP_CHECK(hipMemcpyAsync(d_stage_recs_, h_recs_, (size_t)n * sizeof(Rec),
+                               hipMemcpyHostToDevice, stream_));
+      if (pay_bytes)
+        HIP_CHECK(hipMemcpyAsync(d_stage_pay_, h_pay_, pay_bytes,
+                                 hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
+      const int blocks = (n + 3) / 4;
+      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
+                         d_stage_recs_, d_stage_pay_, n, base, d_hdr_,
+                         d_status_, d_payload_, d_inbox_, d_wpos_, d_by_type_,
+                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
+                         0, stream_, d_bcast_, d_bcast_count_, d_active_,
+                         d_inbox_, d_wpos_, g_);
+    }
+    count_ = base + (u64)n;
+    if (count_ > g_.num_slots)
+      evict_base_ = count_ - g_.num_slots;
+    return base;
+  }
+
+  void sync() {
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipStreamSynchronize(stream_));
+  }
+
+  u32 alloc_bitmap(py::buffer words) {
+    py::buffer_info wi = words.request();
+    if ((size_t)wi.size * wi.itemsize != g_.bitmap_words * sizeof(u64))
+      throw std::invalid_argument("bitmap must be max_agents/64 u64 words");
+    const u32 idx = bitmap_next_++ % g_.num_bitmaps;
+    HIP_CHECK(hipMemcpy(d_bitmaps_ + (size_t)idx * g_.bitmap_words, wi.ptr,
+                        g_.bitmap_words * sizeof(u64), hipMemcpyHostToDevice));
+    return idx;
+  }
+
+  // ---- receive plane ----
+
+  // Returns (counts[n_agents], seqs[n_agents * max_per_agent]) — one
+  // dequeue-kernel launch for the whole poll tick.
+  py::tuple receive_many(py::array_t<u32> agents, int max_per_agent,
+                         bool priority) {
+    const int na = (int)agents.size();
+    if (na == 0)
+      return py::make_tuple(py::array_t<u32>(0), py::array_t<u64>(0));
+    if ((size_t)na * max_per_agent > out_pool_)
+      throw std::invalid_argument("receive batch exceeds output pool");
+    py::array_t<u32> counts(na);
+    py::array_t<u64> seqs((size_t)na * max_per_agent);
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemcpyAsync(d_agents_, agents.data(), na * sizeof(u32),
+                               hipMemcpyHostToDevice, stream_));
+      hipLaunchKernelGGL(k_receive, dim3(na), dim3(256), 0, stream_, d_agents_,
+                         na, max_per_agent, priority ? 1 : 0, evict_base_,
+                         d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
+                         d_carry_, d_carry_n_, d_bitmaps_, d_out_seqs_,
+                         d_out_counts_, d_by_status_, d_received_, g_);
+      HIP_CHECK(hipMemcpyAsync(h_out_counts_, d_out_counts_, na * sizeof(u32),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipMemcpyAsync(h_out_seqs_, d_out_seqs_,
+                               (size_t)na * max_per_agent * sizeof(u64),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    std::memcpy(counts.mutable_data(), h_out_counts_, na * sizeof(u32));
+    std::memcpy(seqs.mutable_data(), h_out_seqs_,
+                (size_t)na * max_per_agent * sizeof(u64));
+    return py::make_tuple(counts, seqs);
+  }
+
+  // ---- message store ----
+
+  // Returns (hdr bytes [n*48], status[n], payload bytes [n*slot_bytes]).
+  py::tuple fetch(py::array_t<u64> seqs) {
+    const int n = (int)seqs.size();
+    if (n == 0)
+      return py::make_tuple(py::bytes(""), py::array_t<u32>(0),
+                            py::bytes(""));
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("fetch batch exceeds staging_batch");
+    {
+      py::gil_scoped_release nogil;
+      // order the gather after any in-flight enqueue on the main stream
+      HIP_CHECK(hipEventRecord(ev_, stream_));
+      HIP_CHECK(hipStreamWaitEvent(copy_stream_, ev_, 0));
+      HIP_CHECK(hipMemcpyAsync(d_seqs_in_, seqs.data(), n * sizeof(u64),
+                               hipMemcpyHostToDevice, copy_stream_));
+      hipLaunchKernelGGL(k_gather, dim3((n + 3) / 4), dim3(256), 0,
+                         copy_stream_, d_seqs_in_, n, d_hdr_, d_status_,
+                         d_payload_, d_fetch_hdr_, d_fetch_status_,
+                         d_fetch_pay_, evict_base_, g_);
+      HIP_CHECK(hipMemcpyAsync(h_fetch_hdr_, d_fetch_hdr_, n * sizeof(Rec),
+                               hipMemcpyDeviceToHost, copy_stream_));
+      HIP_CHECK(hipMemcpyAsync(h_fetch_status_, d_fetch_status_,
+                               n * sizeof(u32), hipMemcpyDeviceToHost,
+                               copy_stream_));
+      HIP_CHECK(hipMemcpyAsync(h_fetch_pay_, d_fetch_pay_,
+                               (size_t)n * g_.slot_bytes,
+                               hipMemcpyDeviceToHost, copy_stream_));
+      HIP_CHECK(hipStreamSynchronize(copy_stream_));
+    }
+    py::array_t<u32> status(n);
+    std::memcpy(status.mutable_data(), h_fetch_status_, n * sizeof(u32));
+    return py::make_tuple(
+        py::bytes(reinterpret_cast<const char *>(h_fetch_hdr_),
+                  (size_t)n * sizeof(Rec)),
+        status,
+        py::bytes(reinterpret_cast<const char *>(h_fetch_pay_),
+                  (size_t)n * g_.slot_bytes));
+  }
+
+  void set_status(u64 seq, u32 st) {
+    hipLaunchKernelGGL(k_set_status, dim3(1), dim3(64), 0, stream_, seq, st,
+                       d_status_, d_by_status_, g_);
+    HIP_CHECK(hipStreamSynchronize(stream_));
+  }
+
+  u32 get_status(u64 seq) {
+    if (seq < evict_base_ || seq >= count_)
+      return ST_DELETED;
+    u32 st;
+    HIP_CHECK(hipMemcpy(&st, d_status_ + (seq % g_.num_slots), sizeof(u32),
+                        hipMemcpyDeviceToHost));
+    return st;
+  }
+
+  // Filter scan over [lo, hi); returns matched seqs (unordered).
+  py::array_t<u64> query_range(u64 lo, u64 hi, int f_sender, int f_receiver,
+                               int f_type, int f_status, double after,
+                               double before, u32 pred_mask, u32 cap) {
+    if (lo < evict_base_)
+      lo = evict_base_;
+    if (hi > count_)
+      hi = count_;
+    if (cap > staging_batch_)
+      cap = staging_batch_;
+    if (lo >= hi)
+      return py::array_t<u64>(0);
+    u32 nmatch = 0;
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemsetAsync(d_match_count_, 0, sizeof(u32), stream_));
+      const u64 span = hi - lo;
+      const int blocks = (int)std::min<u64>((span + 255) / 256, 2048); // grid-stride
+      hipLaunchKernelGGL(k_filter, dim3(blocks), dim3(256), 0, stream_, lo, hi,
+                         f_sender, f_receiver, f_type, f_status, after, before,
+                         pred_mask, d_hdr_, d_status_, d_match_,
+                         d_match_count_, cap, g_);
+      HIP_CHECK(hipMemcpyAsync(&nmatch, d_match_count_, sizeof(u32),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    const u32 take = nmatch < cap ? nmatch : cap;
+    py::array_t<u64> out(take);
+    if (take)
+      HIP_CHECK(hipMemcpy(out.mutable_data(), d_match_, take * sizeof(u64),
+                          hipMemcpyDeviceToHost));
+    return out;
+  }
+
+  py::array_t<u64> search_range(u64 lo, u64 hi, py::bytes needle, bool fold,
+                                u32 cap) {
+    std::string nd = needle;
+    if (nd.empty() || nd.size() > 256)
+      throw std::invalid_argument("needle must be 1..256 bytes");
+    if (lo < evict_base_)
+      lo = evict_base_;
+    if (hi > count_)
+      hi = count_;
+    if (cap > staging_batch_)
+      cap = staging_batch_;
+    if (lo >= hi)
+      return py::array_t<u64>(0);
+    u32 nmatch = 0;
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemcpyAsync(d_needle_, nd.data(), nd.size(),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemsetAsync(d_match_count_, 0, sizeof(u32), stream_));
+      const u64 span = hi - lo;
+      const int blocks = (int)std::min<u64>((span + 3) / 4, 2048);
+      hipLaunchKernelGGL(k_search, dim3(blocks), dim3(256), 0, stream_, lo, hi,
+                         d_needle_, (int)nd.size(), fold ? 1 : 0, d_hdr_,
+                         d_status_, d_payload_, d_match_, d_match_count_, cap,
+                         g_);
+      HIP_CHECK(hipMemcpyAsync(&nmatch, d_match_count_, sizeof(u32),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    const u32 take = nmatch < cap ? nmatch : cap;
+    py::array_t<u64> out(take);
+    if (take)
+      HIP_CHECK(hipMemcpy(out.mutable_data(), d_match_, take * sizeof(u64),
+                          hipMemcpyDeviceToHost));
+    return out;
+  }
+
+  // Inbox window for history listing (peek): returns (wpos, entries).
+  py::tuple inbox_window(u32 agent) {
+    check_agent(agent);
+    ull w;
+    HIP_CHECK(hipMemcpy(&w, d_wpos_ + agent, sizeof(ull),
+                        hipMemcpyDeviceToHost));
+    const ull start = w > g_.inbox_capacity ? w - g_.inbox_capacity : 0;
+    const size_t nn = (size_t)(w - start);
+    py::array_t<u64> out(nn);
+    if (nn) {
+      // the ring window may wrap; copy in up to two pieces
+      std::vector<u64> tmp(nn);
+      const u64 *base = d_inbox_ + (u64)agent * g_.inbox_capacity;
+      const ull s_mod = start % g_.inbox_capacity;
+      const size_t first = std::min<size_t>(nn, g_.inbox_capacity - s_mod);
+      HIP_CHECK(hipMemcpy(tmp.data(), base + s_mod, first * sizeof(u64),
+                          hipMemcpyDeviceToHost));
+      if (first < nn)
+        HIP_CHECK(hipMemcpy(tmp.data() + first, base, (nn - first) * sizeof(u64),
+                            hipMemcpyDeviceToHost));
+      std::memcpy(out.mutable_data(), tmp.data(), nn * sizeof(u64));
+    }
+    return py::make_tuple((u64)w, out);
+  }
+
+  py::array_t<u32> unread_counts(py::array_t<u32> agents) {
+    const int na = (int)agents.size();
+    py::array_t<u32> out(na);
+    if (na == 0)
+      return out;
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemcpyAsync(d_agents_, agents.data(), na * sizeof(u32),
+                               hipMemcpyHostToDevice, stream_));
+      hipLaunchKernelGGL(k_unread, dim3(na), dim3(256), 0, stream_, d_agents_,
+                         na, evict_base_, d_status_, d_inbox_, d_wpos_,
+                         d_unread_, g_);
+      HIP_CHECK(hipMemcpyAsync(h_out_counts_, d_unread_, na * sizeof(u32),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    std::memcpy(out.mutable_data(), h_out_counts_, na * sizeof(u32));
+    return out;
+  }
+
+  // ---- counters / stats ----
+
+  py::dict counters() {
+    py::array_t<u64> by_type(N_TYPES), by_status(N_STATUS);
+    py::array_t<u64> sent(g_.max_agents), received(g_.max_agents);
+    HIP_CHECK(hipMemcpy(by_type.mutable_data(), d_by_type_,
+                        N_TYPES * sizeof(ull), hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(by_status.mutable_data(), d_by_status_,
+                        N_STATUS * sizeof(ull), hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(sent.mutable_data(), d_sent_,
+                        g_.max_agents * sizeof(ull), hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(received.mutable_data(), d_received_,
+                        g_.max_agents * sizeof(ull), hipMemcpyDeviceToHost));
+    py::dict d;
+    d["by_type"] = by_type;
+    d["by_status"] = by_status;
+    d["sent"] = sent;
+    d["received"] = received;
+    return d;
+  }
+
+  // ---- load balancer ----
+
+  void backend_add_load(int idx, long long delta) {
+    hipLaunchKernelGGL(k_add_load, dim3(1), dim3(64), 0, stream_, idx, delta,
+                       d_backend_loads_);
+    HIP_CHECK(hipStreamSynchronize(stream_));
+  }
+
+  py::array_t<i64> backend_loads() {
+    py::array_t<i64> out(g_.num_backends);
+    HIP_CHECK(hipMemcpy(out.mutable_data(), d_backend_loads_,
+                        g_.num_backends * sizeof(ull), hipMemcpyDeviceToHost));
+    return out;
+  }
+
+  // Exact sequential least-loaded dispatch of `requests` requests;
+  // returns the chosen backend per request.
+  py::array_t<u32> lb_dispatch(int requests, int n_backends) {
+    if (n_backends <= 0 || n_backends > 64)
+      throw std::invalid_argument("1..64 backends supported");
+    if ((u32)requests > staging_batch_)
+      throw std::invalid_argument("too many requests per dispatch batch");
+    py::array_t<u32> out(requests);
+    {
+      py::gil_scoped_release nogil;
+      hipLaunchKernelGGL(k_lb_batch, dim3(1), dim3(64), 0, stream_, requests,
+                         n_backends, d_backend_loads_, d_choices_);
+      HIP_CHECK(hipMemcpyAsync(h_choices_, d_choices_,
+                               requests * sizeof(u32), hipMemcpyDeviceToHost,
+                               stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    std::memcpy(out.mutable_data(), h_choices_, requests * sizeof(u32));
+    return out;
+  }
+
+  // ---- accessors ----
+  u64 total_messages() const { return count_; }
+  u64 evict_base() const { return evict_base_; }
+  u32 staging_batch() const { return staging_batch_; }
+  u32 slot_bytes() const { return g_.slot_bytes; }
+  u32 recv_window() const { return g_.recv_window; }
+
+private:
+  void check_agent(u32 idx) const {
+    if (idx >= g_.max_agents)
+      throw std::out_of_range("agent index out of range");
+  }
+
+  void ensure_stage_pay(size_t bytes) {
+    if (bytes <= stage_pay_bytes_)
+      return;
+    size_t nb = stage_pay_bytes_;
+    while (nb < bytes)
+      nb *= 2;
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    HIP_CHECK(hipFree(d_stage_pay_));
+    (void)hipHostFree(h_pay_);
+    HIP_CHECK(hipMalloc(&d_stage_pay_, nb));
+    HIP_CHECK(hipHostMalloc(&h_pay_, nb));
+    stage_pay_bytes_ = nb;
+  }
+
+  QueueGeom g_;
+  int device_;
+  u32 staging_batch_;
+  u64 count_ = 0;
+  u64 evict_base_ = 0;
+  u32 bitmap_next_ = 0;
+  bool released_ = false;
+  size_t out_pool_ = 0;
+  size_t stage_pay_bytes_ = 0;
+
+  hipStream_t stream_{}, copy_stream_{};
+  hipEvent_t ev_{};
+
+  Rec *d_hdr_{};
+  u32 *d_status_{};
+  u8 *d_payload_{};
+  u64 *d_inbox_{};
+  ull *d_wpos_{};
+  ull *d_rpos_{};
+  u64 *d_carry_{};
+  u32 *d_carry_n_{};
+  u32 *d_active_{};
+  u64 *d_bitmaps_{};
+  ull *d_by_type_{};
+  ull *d_by_status_{};
+  ull *d_sent_{};
+  ull *d_received_{};
+  u64 *d_bcast_{};
+  u32 *d_bcast_count_{};
+  ull *d_backend_loads_{};
+  u32 *d_choices_{};
+  u64 *d_match_{};
+  u32 *d_match_count_{};
+  u8 *d_needle_{};
+  u32 *d_agents_{};
+  u32 *d_unread_{};
+  u64 *d_out_seqs_{};
+  u32 *d_out_counts_{};
+  Rec *d_stage_recs_{};
+  u8 *d_stage_pay_{};
+  u64 *d_seqs_in_{};
+  Rec *d_fetch_hdr_{};
+  u32 *d_fetch_status_{};
+  u8 *d_fetch_pay_{};
+
+  Rec *h_recs_{};
+  u8 *h_pay_{};
+  u64 *h_out_seqs_{};
+  u32 *h_out_counts_{};
+  Rec *h_fetch_hdr_{};
+  u32 *h_fetch_status_{};
+  u8 *h_fetch_pay_{};
+  u32 *h_choices_{};
+};
+
+// ---------------------------------------------------------------------------
+// python module
+// ---------------------------------------------------------------------------
+
+PYBIND11_MODULE(_swarmq, m) {
+  m.doc() = "MI355X-native GPU message-queue engine (CDNA4 HIP kernels)";
+  m.attr("RECV_WINDOW") = RECV_WINDOW;
+
+  m.def("device_count", []() {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess)
+      return 0;
+    return n;
+  });
+
+  py::class_<DeviceQueue>(m, "DeviceQueue")
+      .def(py::init<u32, u32, u32, u32, u32, u32, u32, int>(),
+           py::arg("num_slots"), py::arg("slot_bytes"), py::arg("max_agents"),
+           py::arg("inbox_capacity"), py::arg("num_bitmaps"),
+           py::arg("num_backends"), py::arg("staging_batch"),
+           py::arg("device") = 0)
+      .def("register_agent", &DeviceQueue::register_agent)
+      .def("deregister_agent", &DeviceQueue::deregister_agent)
+      .def("active_agents", &DeviceQueue::active_agents)
+      .def("enqueue_batch", &DeviceQueue::enqueue_batch)
+      .def("enqueue_batch_async", &DeviceQueue::enqueue_batch_async)
+      .def("sync", &DeviceQueue::sync)
+      .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
+      .def("receive_many", &DeviceQueue::receive_many)
+      .def("fetch", &DeviceQueue::fetch)
+      .def("set_status", &DeviceQueue::set_status)
+      .def("get_status", &DeviceQueue::get_status)
+      .def("query_range", &DeviceQueue::query_range)
+      .def("search_range", &DeviceQueue::search_range)
+      .def("inbox_window", &DeviceQueue::inbox_window)
+      .def("unread_counts", &DeviceQueue::unread_counts)
+      .def("counters", &DeviceQueue::counters)
+      .def("backend_add_load", &DeviceQueue::backend_add_load)
+      .def("backend_loads", &DeviceQueue::backend_loads)
+      .def("lb_dispatch", &DeviceQueue::lb_dispatch)
+      .def("total_messages", &DeviceQueue::total_messages)
+      .def("evict_base", &DeviceQueue::evict_base)
+      .def("staging_batch", &DeviceQueue::staging_batch)
+      .def("slot_bytes", &DeviceQueue::slot_bytes)
+      .def("recv_window", &DeviceQueue::recv_window)
+      .def("release", &DeviceQueue::release);
+}

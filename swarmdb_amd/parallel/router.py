@@ -1,0 +1,160 @@
+"""Cross-GPU message routing — RCCL all-to-all over xGMI.
+
+Agents are sharded across the node's GPUs (one process per GPU,
+``torch.distributed`` with the nccl backend = RCCL on ROCm); each tick,
+every rank partitions its outbound batch by destination rank and the
+node exchanges (records, payloads) with a single
+``all_to_all_single`` — direct point-to-point traffic over the
+fully-connected xGMI mesh (7 links x ~153 GB/s per GPU), which is exactly
+the shape all-to-all wants (no ring serialization). Broadcast messages
+are replicated to every rank; each rank fans out to its locally resident
+agents only.
+
+This replaces the reference's single shared Kafka broker as the
+cross-process transport (SURVEY.md §2.4 rows "partitioned topic" /
+"cross-process shared state"; BASELINE config 4).
+
+Works on CPU tensors with the gloo backend for CI without GPUs.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..runtime.engine import BROADCAST, REC_DTYPE
+
+REC_BYTES = REC_DTYPE.itemsize  # 48
+
+
+def shard_of(receiver_idx: np.ndarray, world_size: int) -> np.ndarray:
+    """Owner rank of each (global) receiver index. Round-robin keeps
+    every rank's agent set contiguous-in-modulus and load-balanced; the
+    facade's string-keyed path uses the stable FNV hash instead
+    (utils/hashing.shard_for)."""
+    return receiver_idx % world_size
+
+
+class CrossGpuRouter:
+    """One all-to-all exchange per tick.
+
+    Wire format per destination rank: n records (48 B each, payload_off
+    rebased to that destination's payload chunk) followed by the payload
+    bytes. Counts go first in a small all-to-all so receive buffers can
+    be sized exactly.
+    """
+
+    def __init__(self, device: torch.device, group: Optional[object] = None):
+        self.device = device
+        self.group = group
+        self.world = dist.get_world_size(group)
+        self.rank = dist.get_rank(group)
+
+    def route(
+        self, recs: np.ndarray, payloads: bytes
+    ) -> Tuple[np.ndarray, bytes]:
+        """Partition by destination, exchange, return this rank's inbound
+        (records, payloads) — remote plus own, payload offsets rebased to
+        the returned buffer."""
+        W = self.world
+        recv = recs["receiver"]
+        dest = np.where(
+            recv == BROADCAST, self.rank, recv % np.uint32(W)
+        ).astype(np.int64)
+        bmask = recv == BROADCAST
+
+        # build per-destination chunks (broadcasts replicated to all)
+        src = np.frombuffer(payloads, dtype=np.uint8)
+        all_lens = recs["payload_len"].astype(np.int64)
+        all_offs = recs["payload_off"].astype(np.int64)
+        # fast path: uniform 16-B-aligned stride (the batch hot path) —
+        # payload gather becomes a vectorized 2-D fancy index
+        stride = int(((all_lens.max() if len(recs) else 0) + 15) // 16 * 16)
+        uniform = (
+            len(recs) > 0
+            and (all_lens == all_lens[0]).all()
+            and (all_offs == np.arange(len(recs), dtype=np.int64) * stride).all()
+            and len(src) >= len(recs) * stride
+        )
+        out_blobs = []
+        for d in range(W):
+            sel = (dest == d) | bmask
+            sub = recs[sel].copy()
+            if uniform:
+                pay_bytes = (
+                    src[: len(recs) * stride]
+                    .reshape(len(recs), stride)[sel]
+                    .tobytes()
+                )
+                sub["payload_off"] = (
+                    np.arange(len(sub), dtype=np.uint64) * np.uint64(stride)
+                )
+            else:
+                lens16 = ((sub["payload_len"].astype(np.int64) + 15) // 16) * 16
+                offs = np.zeros(len(sub), dtype=np.int64)
+                np.cumsum(lens16[:-1], out=offs[1:])
+                dst = np.zeros(int(lens16.sum()), dtype=np.uint8)
+                for i in range(len(sub)):
+                    o, l, no = (
+                        int(sub["payload_off"][i]),
+                        int(sub["payload_len"][i]),
+                        int(offs[i]),
+                    )
+                    dst[no : no + l] = src[o : o + l]
+                sub["payload_off"] = offs.astype(np.uint64)
+                pay_bytes = dst.tobytes()
+            out_blobs.append(sub.tobytes() + pay_bytes)
+
+        sizes = torch.tensor(
+            [len(b) for b in out_blobs], dtype=torch.int64, device=self.device
+        )
+        nmsgs = torch.tensor(
+            [int(((dest == d) | bmask).sum()) for d in range(W)],
+            dtype=torch.int64,
+            device=self.device,
+        )
+        in_sizes = torch.empty_like(sizes)
+        in_nmsgs = torch.empty_like(nmsgs)
+        dist.all_to_all_single(in_sizes, sizes, group=self.group)
+        dist.all_to_all_single(in_nmsgs, nmsgs, group=self.group)
+
+        send_buf = torch.frombuffer(
+            bytearray(b"".join(out_blobs)), dtype=torch.uint8
+        ).to(self.device, non_blocking=False)
+        total_in = int(in_sizes.sum().item())
+        recv_buf = torch.empty(total_in, dtype=torch.uint8, device=self.device)
+        dist.all_to_all_single(
+            recv_buf,
+            send_buf,
+            output_split_sizes=in_sizes.tolist(),
+            input_split_sizes=sizes.tolist(),
+            group=self.group,
+        )
+        raw = recv_buf.cpu().numpy().tobytes()
+
+        # unpack: concatenate records, rebase payload offsets
+        rec_parts, pay_parts = [], []
+        pay_base = 0
+        off = 0
+        for d in range(W):
+            sz = int(in_sizes[d].item())
+            nm = int(in_nmsgs[d].item())
+            blob = raw[off : off + sz]
+            off += sz
+            rr = np.frombuffer(
+                blob[: nm * REC_BYTES], dtype=REC_DTYPE, count=nm
+            ).copy()
+            pp = blob[nm * REC_BYTES :]
+            rr["payload_off"] += np.uint64(pay_base)
+            pay_base += len(pp)
+            rec_parts.append(rr)
+            pay_parts.append(pp)
+        all_recs = (
+            np.concatenate(rec_parts)
+            if rec_parts
+            else np.empty(0, dtype=REC_DTYPE)
+        )
+        return all_recs, b"".join(pay_parts)

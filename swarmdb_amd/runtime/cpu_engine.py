@@ -84,6 +84,11 @@ class CpuEngine(Engine):
         self._active = np.zeros(c.max_agents, dtype=bool)
         self._inbox: Dict[int, _U64Ring] = {}
         self._cursor = np.zeros(c.max_agents, dtype=np.int64)
+        # visible-but-unconsumed entries from partial drains (the inbox
+        # ring itself is append-only history; the GPU engine keeps the
+        # same per-agent carry buffer device-side)
+        self._carry: Dict[int, np.ndarray] = {}
+        self._window = 4096  # max entries examined per receive call
 
         self._bitmaps: List[np.ndarray] = []
 
@@ -211,8 +216,23 @@ class CpuEngine(Engine):
             if ring is None:
                 return np.empty(0, dtype=np.uint64)
             cur = int(self._cursor[agent_idx])
-            pend = ring.view()[cur:]
+            carry = self._carry.get(agent_idx)
+            ncarry = 0 if carry is None else len(carry)
+            # examine at most `window` entries per call (the GPU dequeue
+            # kernel's LDS window); the cursor advances past what was
+            # examined, leftovers go to the carry buffer
+            take_new = min(ring.n - cur, self._window - ncarry)
+            fresh = ring.view()[cur : cur + take_new]
+            self._cursor[agent_idx] = cur + take_new
+            if ncarry:
+                pend = np.sort(np.concatenate((carry, fresh)))
+            else:
+                # sort into seq order regardless of append order: batched
+                # enqueue appends per-receiver groups (and the GPU engine
+                # appends with atomics) — the dequeue kernel sorts in LDS
+                pend = np.sort(fresh)
             if len(pend) == 0:
+                self._carry.pop(agent_idx, None)
                 return pend.copy()
             ok = self._visible_mask(agent_idx, pend)
             cand = pend[ok]
@@ -221,22 +241,11 @@ class CpuEngine(Engine):
                 # stable sort by priority desc (seq order preserved within)
                 cand = cand[np.argsort(-pri, kind="stable")]
             out = cand[:max_messages]
-            if priority_order:
-                # priority mode consumes only what it returns; visible
-                # unconsumed entries stay pending. Invisible entries are
-                # dropped for good (visibility is immutable).
-                taken = np.isin(pend, out)
-                rest = np.sort(pend[~taken & ok])
-                ring.n = cur
-                ring.append_many(rest)
+            rest = cand[max_messages:]
+            if len(rest):
+                self._carry[agent_idx] = np.sort(rest) if priority_order else rest.copy()
             else:
-                if len(cand) > max_messages:
-                    # advance cursor past delivered portion only
-                    last = out[-1]
-                    pos = int(np.flatnonzero(pend == last)[0])
-                    self._cursor[agent_idx] = cur + pos + 1
-                else:
-                    self._cursor[agent_idx] = cur + len(pend)
+                self._carry.pop(agent_idx, None)
             if len(out):
                 prev = self._status[out]
                 was_delivered = prev == ST_DELIVERED
@@ -256,7 +265,7 @@ class CpuEngine(Engine):
             ring = self._inbox.get(agent_idx)
             if ring is None:
                 return np.empty(0, dtype=np.uint64)
-            seqs = ring.view()
+            seqs = np.sort(ring.view())  # log (seq) order
             alive = self._status[seqs] != ST_DELETED
             return seqs[alive].copy()
 

@@ -233,6 +233,18 @@ class Engine(abc.ABC):
         GPU — BASELINE config 5; the mechanism the reference lacks,
         SURVEY.md §2.2 'LLM load balancing')."""
 
+    def dispatch_batch(self, requests: int, n_backends: int) -> np.ndarray:
+        """Exact sequential least-loaded dispatch of `requests` requests:
+        each pick increments the chosen backend's in-flight count. One
+        wavefront-shuffle reduction kernel on GPU; returns the chosen
+        backend index per request."""
+        out = np.empty(requests, dtype=np.uint32)
+        for i in range(requests):
+            b = self.least_loaded_backend(n_backends)
+            self.backend_add_load(b, 1)
+            out[i] = b
+        return out
+
     # --- lifecycle ---
 
     def close(self) -> None:  # noqa: B027

@@ -701,8 +701,7 @@ class SwarmsDB:
                     idx = self._llm_backend_idx[pinned]
                     self.engine.backend_add_load(idx, 1)
                     return pinned
-        idx = self.engine.least_loaded_backend(n)
-        self.engine.backend_add_load(idx, 1)
+        idx = int(self.engine.dispatch_batch(1, n)[0])
         return self._llm_backends[idx]
 
     def complete_llm_request(self, backend_id: str) -> None:

@@ -1,0 +1,353 @@
+"""GPU delivery engine — HBM-resident rings via the _swarmq HIP extension.
+
+Implements the :class:`Engine` contract over ``DeviceQueue``
+(csrc/swarmq_module.hip): the message log, inbox rings, read cursors,
+status words, visibility bitmaps and load counters all live in MI355X
+HBM3E; every hot operation is one H2D staging copy + one CDNA4 kernel.
+
+Fails loudly if the extension is missing or no GPU is visible — there is
+no silent eager fallback (use CpuEngine explicitly for CPU runs).
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..core.config import QueueConfig
+from .engine import (
+    REC_DTYPE,
+    ST_DELETED,
+    ST_DELIVERED,
+    Engine,
+)
+
+_FETCH_DTYPE = np.dtype(REC_DTYPE.descr + [("status", np.uint8), ("seq", np.uint64)])
+
+
+def _load_ext():
+    try:
+        from .. import _swarmq  # type: ignore
+    except ImportError as e:  # pragma: no cover
+        raise RuntimeError(
+            "the _swarmq HIP extension is not built — run "
+            "`python build_ext.py` (hipcc, gfx950). The GPU engine has no "
+            "eager fallback by design."
+        ) from e
+    return _swarmq
+
+
+class GpuEngine(Engine):
+    def __init__(self, config: Optional[QueueConfig] = None):
+        self.cfg = config or QueueConfig()
+        c = self.cfg
+        ext = _load_ext()
+        if ext.device_count() == 0:  # pragma: no cover
+            raise RuntimeError(
+                "no HIP device visible — GpuEngine requires an MI355X"
+            )
+        if c.max_agents % 64 != 0:
+            raise ValueError("max_agents must be a multiple of 64")
+        self._lock = threading.RLock()
+        self.q = ext.DeviceQueue(
+            num_slots=c.num_slots,
+            slot_bytes=c.slot_bytes,
+            max_agents=c.max_agents,
+            inbox_capacity=c.inbox_capacity,
+            num_bitmaps=4096,
+            num_backends=c.num_backends,
+            staging_batch=c.staging_batch,
+            device=c.device_index,
+        )
+        self._staging = c.staging_batch
+        self._slot_bytes = c.slot_bytes
+        # host-side receive-timestamp windows (processing_rate probe);
+        # counts come from the dequeue kernel's out_counts
+        self._recv_ts: Dict[int, List[Tuple[float, int]]] = {}
+
+    # --- registry ---
+
+    def register_agent(self, agent_idx: int) -> None:
+        self.q.register_agent(agent_idx)
+
+    def deregister_agent(self, agent_idx: int) -> None:
+        self.q.deregister_agent(agent_idx)
+
+    def active_agents(self) -> np.ndarray:
+        return self.q.active_agents().astype(bool)
+
+    # --- send plane ---
+
+    def _pack_aligned(self, recs: np.ndarray, payloads: bytes):
+        """Ensure 16-B aligned staging offsets (the enqueue kernel copies
+        uint4 chunks). Repacks only when the caller's offsets aren't
+        aligned already."""
+        offs = recs["payload_off"]
+        lens = recs["payload_len"]
+        if len(recs) and (offs % 16 == 0).all():
+            slack = (np.int64(len(payloads)) - (offs.astype(np.int64) + lens)).min() if len(recs) else 0
+            # kernel may over-read up to the 16B round-up of each payload
+            pad_needed = int(((lens + 15) // 16 * 16 - lens).max()) if len(recs) else 0
+            if slack >= pad_needed:
+                return recs, payloads
+        aligned_lens = (lens.astype(np.int64) + 15) // 16 * 16
+        new_offs = np.zeros(len(recs), dtype=np.int64)
+        np.cumsum(aligned_lens[:-1], out=new_offs[1:])
+        buf = np.zeros(int(aligned_lens.sum()) + 16, dtype=np.uint8)
+        src = np.frombuffer(payloads, dtype=np.uint8)
+        for i in range(len(recs)):
+            o, l, no = int(offs[i]), int(lens[i]), int(new_offs[i])
+            buf[no : no + l] = src[o : o + l]
+        out = recs.copy()
+        out["payload_off"] = new_offs.astype(np.uint64)
+        return out, buf.tobytes()
+
+    def enqueue_batch(self, recs: np.ndarray, payloads: bytes) -> np.ndarray:
+        n = len(recs)
+        if n == 0:
+            return np.empty(0, dtype=np.uint64)
+        recs, payloads = self._pack_aligned(np.ascontiguousarray(recs), payloads)
+        with self._lock:
+            seqs = np.empty(n, dtype=np.uint64)
+            done = 0
+            while done < n:
+                chunk = min(self._staging, n - done)
+                sub = recs[done : done + chunk]
+                lo = int(sub["payload_off"][0])
+                hi = int(sub["payload_off"][-1] + sub["payload_len"][-1])
+                sub = sub.copy()
+                sub["payload_off"] -= np.uint64(lo)
+                base = self.q.enqueue_batch(
+                    sub.tobytes(), payloads[lo : hi + 16], chunk
+                )
+                seqs[done : done + chunk] = np.arange(
+                    base, base + chunk, dtype=np.uint64
+                )
+                done += chunk
+            return seqs
+
+    def alloc_bitmap(self, bits: np.ndarray) -> int:
+        words = np.packbits(
+            bits.astype(bool), bitorder="little"
+        ).view(np.uint64)
+        return int(self.q.alloc_bitmap(words.tobytes()))
+
+    # --- receive plane ---
+
+    def receive(
+        self, agent_idx: int, max_messages: int, priority_order: bool = False
+    ) -> np.ndarray:
+        counts, seqs = self.receive_many(
+            np.array([agent_idx], dtype=np.uint32), max_messages, priority_order
+        )
+        return seqs[: counts[0]]
+
+    def receive_many(
+        self,
+        agent_idxs: np.ndarray,
+        max_per_agent: int,
+        priority_order: bool = False,
+    ) -> Tuple[np.ndarray, np.ndarray]:
+        agent_idxs = np.ascontiguousarray(agent_idxs, dtype=np.uint32)
+        with self._lock:
+            counts, flat = self.q.receive_many(
+                agent_idxs, int(max_per_agent), bool(priority_order)
+            )
+        counts = counts.astype(np.int64)
+        now = time.time()
+        out_chunks = []
+        for i, a in enumerate(agent_idxs):
+            c = int(counts[i])
+            if c:
+                out_chunks.append(
+                    flat[i * max_per_agent : i * max_per_agent + c]
+                )
+                win = self._recv_ts.setdefault(int(a), [])
+                win.append((now, c))
+                if len(win) > 4096:
+                    del win[:2048]
+        seqs = (
+            np.concatenate(out_chunks)
+            if out_chunks
+            else np.empty(0, dtype=np.uint64)
+        )
+        return counts, seqs
+
+    def peek_inbox(self, agent_idx: int) -> np.ndarray:
+        _, entries = self.q.inbox_window(int(agent_idx))
+        entries = np.asarray(entries, dtype=np.uint64)
+        if len(entries) == 0:
+            return entries
+        entries = np.sort(entries)
+        eb = self.q.evict_base()
+        entries = entries[entries >= eb]
+        if len(entries) == 0:
+            return entries
+        # drop tombstoned
+        st = self._statuses(entries)
+        return entries[st != ST_DELETED]
+
+    def _statuses(self, seqs: np.ndarray) -> np.ndarray:
+        hdrs, _ = self.fetch(seqs)
+        return hdrs["status"]
+
+    def unread_count(self, agent_idx: int) -> int:
+        out = self.q.unread_counts(np.array([agent_idx], dtype=np.uint32))
+        return int(out[0])
+
+    # --- message store ---
+
+    def fetch(self, seqs: np.ndarray) -> Tuple[np.ndarray, List[bytes]]:
+        seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
+        n = len(seqs)
+        out = np.zeros(n, dtype=_FETCH_DTYPE)
+        pays: List[bytes] = []
+        done = 0
+        while done < n:
+            chunk = min(self._staging, n - done)
+            sub = seqs[done : done + chunk]
+            hdr_b, status, pay_b = self.q.fetch(sub)
+            hdr = np.frombuffer(hdr_b, dtype=REC_DTYPE, count=chunk)
+            for name in REC_DTYPE.names:
+                out[name][done : done + chunk] = hdr[name]
+            out["status"][done : done + chunk] = status.astype(np.uint8)
+            out["seq"][done : done + chunk] = sub
+            sb = self._slot_bytes
+            lens = hdr["payload_len"]
+            pays.extend(
+                pay_b[i * sb : i * sb + int(lens[i])] for i in range(chunk)
+            )
+            done += chunk
+        return out, pays
+
+    def set_status(self, seq: int, status: int) -> None:
+        self.q.set_status(int(seq), int(status))
+
+    def get_status(self, seq: int) -> int:
+        return int(self.q.get_status(int(seq)))
+
+    def query(
+        self,
+        sender: Optional[int] = None,
+        receiver: Optional[int] = None,
+        type_code: Optional[int] = None,
+        status: Optional[int] = None,
+        after: Optional[float] = None,
+        before: Optional[float] = None,
+        limit: int = 100,
+    ) -> np.ndarray:
+        mask = 0
+        if sender is not None:
+            mask |= 1
+        if receiver is not None:
+            mask |= 2
+        if type_code is not None:
+            mask |= 4
+        if status is not None:
+            mask |= 8
+        if after is not None:
+            mask |= 16
+        if before is not None:
+            mask |= 32
+        limit = min(int(limit), self._staging)
+        total = self.q.total_messages()
+        eb = self.q.evict_base()
+        found: List[np.ndarray] = []
+        nfound = 0
+        hi = total
+        chunk = max(1 << 20, limit)
+        # scan newest chunks first until the limit fills
+        while hi > eb and nfound < limit:
+            lo = max(eb, hi - chunk)
+            seqs = self.q.query_range(
+                lo,
+                hi,
+                -1 if sender is None else int(sender),
+                -1 if receiver is None else int(receiver),
+                -1 if type_code is None else int(type_code),
+                -1 if status is None else int(status),
+                0.0 if after is None else float(after),
+                0.0 if before is None else float(before),
+                mask,
+                self._staging,
+            )
+            seqs = np.sort(np.asarray(seqs, dtype=np.uint64))[::-1]
+            found.append(seqs)
+            nfound += len(seqs)
+            hi = lo
+        if not found:
+            return np.empty(0, dtype=np.uint64)
+        return np.concatenate(found)[:limit]
+
+    def search(self, needle: bytes, case_sensitive: bool, limit: int) -> np.ndarray:
+        limit = min(int(limit), self._staging)
+        total = self.q.total_messages()
+        eb = self.q.evict_base()
+        found: List[np.ndarray] = []
+        nfound = 0
+        hi = total
+        chunk = 1 << 20
+        while hi > eb and nfound < limit:
+            lo = max(eb, hi - chunk)
+            seqs = self.q.search_range(
+                lo, hi, needle, not case_sensitive, self._staging
+            )
+            seqs = np.sort(np.asarray(seqs, dtype=np.uint64))[::-1]
+            found.append(seqs)
+            nfound += len(seqs)
+            hi = lo
+        if not found:
+            return np.empty(0, dtype=np.uint64)
+        return np.concatenate(found)[:limit]
+
+    def delete(self, seq: int) -> bool:
+        if self.get_status(seq) == ST_DELETED:
+            return False
+        self.set_status(int(seq), ST_DELETED)
+        return True
+
+    # --- counters / stats ---
+
+    def total_messages(self) -> int:
+        return int(self.q.total_messages())
+
+    def stats_arrays(self) -> Dict[str, np.ndarray]:
+        c = self.q.counters()
+        return {
+            "by_type": np.asarray(c["by_type"], dtype=np.int64),
+            "by_status": np.asarray(c["by_status"], dtype=np.int64),
+            "sent": np.asarray(c["sent"], dtype=np.int64),
+            "received": np.asarray(c["received"], dtype=np.int64),
+        }
+
+    def recv_rate_window(self, agent_idx: int, window_s: float) -> int:
+        win = self._recv_ts.get(int(agent_idx), [])
+        cutoff = time.time() - window_s
+        return sum(c for t, c in win if t >= cutoff)
+
+    # --- load balancer ---
+
+    def backend_add_load(self, backend_idx: int, delta: int) -> None:
+        self.q.backend_add_load(int(backend_idx), int(delta))
+
+    def backend_loads(self) -> np.ndarray:
+        return np.asarray(self.q.backend_loads(), dtype=np.int64)
+
+    def least_loaded_backend(self, n_backends: int) -> int:
+        loads = self.backend_loads()[:n_backends]
+        return int(np.argmin(loads))
+
+    def dispatch_batch(self, requests: int, n_backends: int) -> np.ndarray:
+        """Batched exact least-loaded dispatch — the wavefront min-reduce
+        kernel (BASELINE config 5)."""
+        return np.asarray(
+            self.q.lb_dispatch(int(requests), int(n_backends)), dtype=np.uint32
+        )
+
+    # --- lifecycle ---
+
+    def close(self) -> None:
+        self.q.release()
