@@ -168,7 +168,12 @@ def main() -> int:
         )
         ndel = int(counts.sum())
         if not args.no_gather and ndel:
-            engine.fetch(seqs)  # payload gather + D2H: bytes land on host
+            # payload gather + D2H: bytes land in pinned host memory
+            deliver = getattr(engine, "deliver_payloads", None)
+            if deliver is not None:
+                deliver(seqs)
+            else:
+                engine.fetch(seqs)
         sent_total += sent_local
         recv_total += ndel
         return ndel

@@ -64,9 +64,9 @@ class GpuEngine(Engine):
         )
         self._staging = c.staging_batch
         self._slot_bytes = c.slot_bytes
-        # host-side receive-timestamp windows (processing_rate probe);
-        # counts come from the dequeue kernel's out_counts
-        self._recv_ts: Dict[int, List[Tuple[float, int]]] = {}
+        # host-side receive-event log (processing_rate probe); one entry
+        # per poll tick, pruned as it grows
+        self._recv_events: List[Tuple[float, np.ndarray, np.ndarray]] = []
 
     # --- registry ---
 
@@ -110,6 +110,7 @@ class GpuEngine(Engine):
         if n == 0:
             return np.empty(0, dtype=np.uint64)
         recs, payloads = self._pack_aligned(np.ascontiguousarray(recs), payloads)
+        pay_view = np.frombuffer(payloads, dtype=np.uint8)
         with self._lock:
             seqs = np.empty(n, dtype=np.uint64)
             done = 0
@@ -118,10 +119,12 @@ class GpuEngine(Engine):
                 sub = recs[done : done + chunk]
                 lo = int(sub["payload_off"][0])
                 hi = int(sub["payload_off"][-1] + sub["payload_len"][-1])
-                sub = sub.copy()
-                sub["payload_off"] -= np.uint64(lo)
+                if done or lo:
+                    sub = sub.copy()
+                    sub["payload_off"] -= np.uint64(lo)
+                # numpy views via the buffer protocol: no bytes copies
                 base = self.q.enqueue_batch(
-                    sub.tobytes(), payloads[lo : hi + 16], chunk
+                    sub, pay_view[lo : min(hi + 16, len(pay_view))], chunk
                 )
                 seqs[done : done + chunk] = np.arange(
                     base, base + chunk, dtype=np.uint64
@@ -157,23 +160,17 @@ class GpuEngine(Engine):
                 agent_idxs, int(max_per_agent), bool(priority_order)
             )
         counts = counts.astype(np.int64)
-        now = time.time()
-        out_chunks = []
-        for i, a in enumerate(agent_idxs):
-            c = int(counts[i])
-            if c:
-                out_chunks.append(
-                    flat[i * max_per_agent : i * max_per_agent + c]
-                )
-                win = self._recv_ts.setdefault(int(a), [])
-                win.append((now, c))
-                if len(win) > 4096:
-                    del win[:2048]
-        seqs = (
-            np.concatenate(out_chunks)
-            if out_chunks
-            else np.empty(0, dtype=np.uint64)
-        )
+        total = int(counts.sum())
+        if total == 0:
+            return counts, np.empty(0, dtype=np.uint64)
+        # vectorized compaction of the dense [n_agents, K] output
+        mat = flat.reshape(len(agent_idxs), max_per_agent)
+        mask = np.arange(max_per_agent)[None, :] < counts[:, None]
+        seqs = mat[mask]
+        # O(1)-per-tick receive-rate bookkeeping (processing_rate probe)
+        self._recv_events.append((time.time(), agent_idxs, counts))
+        if len(self._recv_events) > 1024:
+            del self._recv_events[:512]
         return counts, seqs
 
     def peek_inbox(self, agent_idx: int) -> np.ndarray:
@@ -222,6 +219,20 @@ class GpuEngine(Engine):
             )
             done += chunk
         return out, pays
+
+    def deliver_payloads(self, seqs: np.ndarray) -> int:
+        """Gather + D2H the payloads of `seqs` into pinned host memory
+        (the delivery step of the hot path) without building per-message
+        Python objects. Returns bytes landed."""
+        seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
+        n = len(seqs)
+        done = 0
+        total = 0
+        while done < n:
+            chunk = min(self._staging, n - done)
+            total += int(self.q.fetch_raw(seqs[done : done + chunk]))
+            done += chunk
+        return total
 
     def set_status(self, seq: int, status: int) -> None:
         self.q.set_status(int(seq), int(status))
@@ -324,9 +335,15 @@ class GpuEngine(Engine):
         }
 
     def recv_rate_window(self, agent_idx: int, window_s: float) -> int:
-        win = self._recv_ts.get(int(agent_idx), [])
         cutoff = time.time() - window_s
-        return sum(c for t, c in win if t >= cutoff)
+        total = 0
+        for t, agents, counts in reversed(self._recv_events):
+            if t < cutoff:
+                break
+            hit = counts[agents == agent_idx]
+            if len(hit):
+                total += int(hit.sum())
+        return total
 
     # --- load balancer ---
 

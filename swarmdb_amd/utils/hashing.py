@@ -33,9 +33,17 @@ def partition_for(agent_id: str, num_partitions: int) -> int:
     return stable_hash(agent_id) % max(1, num_partitions)
 
 
+def mix64(h: int) -> int:
+    """splitmix64 finalizer — FNV-1a's raw bits distribute poorly under
+    small moduli on short keys; this avalanches them."""
+    h = (h ^ (h >> 30)) * 0xBF58476D1CE4E5B9 & _MASK64
+    h = (h ^ (h >> 27)) * 0x94D049BB133111EB & _MASK64
+    return h ^ (h >> 31)
+
+
 def shard_for(agent_id: str, world_size: int) -> int:
-    """agent -> GPU rank for cross-GPU sharding (uses the high bits so it
-    decorrelates from partition_for's low-bit modulo)."""
+    """agent -> GPU rank for cross-GPU sharding (mixed so it decorrelates
+    from partition_for's low-bit modulo)."""
     if world_size <= 1:
         return 0
-    return (stable_hash(agent_id) >> 32) % world_size
+    return mix64(stable_hash(agent_id)) % world_size
