@@ -171,7 +171,7 @@ def main() -> int:
             # payload gather + D2H: bytes land in pinned host memory
             deliver = getattr(engine, "deliver_payloads", None)
             if deliver is not None:
-                deliver(seqs)
+                deliver(seqs, args.payload)
             else:
                 engine.fetch(seqs)
         sent_total += sent_local

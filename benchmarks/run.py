@@ -1,0 +1,273 @@
+#!/usr/bin/env python3
+"""The five BASELINE.json benchmark configs.
+
+  1 cpu-loopback   single-agent register + send/receive loopback,
+                   in-process CPU queue (plumbing)
+  2 p2p-gpu        2 agents point-to-point, GPU ring on 1 MI355X, 1 KB msgs
+  3 group-fanout   256 agents / 8 groups, group-broadcast fan-out on
+                   1 MI355X, priority-sort dequeue
+  4 alltoall       4096 agents sharded across N GPUs, RCCL all-to-all
+                   (launch under torch.distributed.run; this is bench.py
+                   with --agents 4096/N)
+  5 loadbalancer   8 mock backends, least-loaded dispatch + JSON history
+                   spill
+
+Usage:
+  python -m benchmarks.run --config {1,2,3,5} [--seconds S]
+  python -m torch.distributed.run --nproc-per-node N benchmarks/run.py --config 4
+
+Each config prints one JSON line:
+  {"config": n, "name": ..., "messages_per_s": ..., "p50_latency_ms": ...}
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import numpy as np
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+from swarmdb_amd import MessagePriority, QueueConfig, SwarmsDB  # noqa: E402
+from swarmdb_amd.runtime.engine import (  # noqa: E402
+    NO_BITMAP,
+    REC_DTYPE,
+    VIS_ALL,
+)
+
+
+def _emit(config: int, name: str, msgs: int, elapsed: float, lat_ms: float,
+          extra=None):
+    out = {
+        "config": config,
+        "name": name,
+        "messages_per_s": round(msgs / elapsed, 1),
+        "p50_latency_ms": round(lat_ms, 4),
+        "messages": msgs,
+        "seconds": round(elapsed, 3),
+    }
+    if extra:
+        out.update(extra)
+    print(json.dumps(out))
+
+
+def _gpu_available() -> bool:
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def config1_cpu_loopback(seconds: float) -> None:
+    """Single agent, in-process CPU queue: the reference-shaped per-message
+    API path (register + send_message + receive_messages loop)."""
+    cfg = QueueConfig(use_gpu=False, auto_save=False, max_agents=64)
+    db = SwarmsDB(config=cfg)
+    db.register_agent("loop")
+    payload = "x" * 1024
+    lat = []
+    n = 0
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        db.send_message("loop", payload, receiver_id="loop")
+        got = db.receive_messages("loop", timeout=0)
+        lat.append(time.perf_counter() - s)
+        assert len(got) == 1
+        n += 1
+    elapsed = time.perf_counter() - t0
+    db.close()
+    _emit(1, "cpu-loopback", n, elapsed, float(np.median(lat) * 1000))
+
+
+def _make_batch(rng, n, senders, receivers, payload_bytes):
+    stride = (payload_bytes + 15) // 16 * 16
+    recs = np.zeros(n, dtype=REC_DTYPE)
+    recs["sender"] = rng.choice(senders, n)
+    recs["receiver"] = rng.choice(receivers, n)
+    recs["priority"] = rng.integers(0, 4, n)
+    recs["timestamp"] = time.time()
+    recs["vis_mode"] = VIS_ALL
+    recs["bitmap"] = NO_BITMAP
+    recs["payload_len"] = payload_bytes
+    recs["content_len"] = payload_bytes
+    recs["payload_off"] = np.arange(n, dtype=np.uint64) * stride
+    payload = rng.integers(32, 127, n * stride, dtype=np.uint8).tobytes()
+    return recs, payload
+
+
+def config2_p2p_gpu(seconds: float) -> None:
+    """2 agents point-to-point over the GPU ring, 1 KB chat messages."""
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    cfg = QueueConfig(use_gpu=True, auto_save=False, max_agents=64,
+                      num_slots=1 << 20, staging_batch=16384)
+    eng = GpuEngine(cfg)
+    eng.register_agent(0)
+    eng.register_agent(1)
+    rng = np.random.default_rng(0)
+    batch = 16384
+    recs, payload = _make_batch(rng, batch, np.array([0]), np.array([1]), 1024)
+    agents = np.array([1], dtype=np.uint32)
+    lat = []
+    n = 0
+    # warmup
+    for _ in range(3):
+        eng.enqueue_batch(recs, payload)
+        eng.receive_many(agents, batch)
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        eng.enqueue_batch(recs, payload)
+        counts, seqs = eng.receive_many(agents, batch)
+        eng.deliver_payloads(seqs, 1024)
+        lat.append(time.perf_counter() - s)
+        n += int(counts.sum())
+    elapsed = time.perf_counter() - t0
+    eng.close()
+    _emit(2, "p2p-gpu-1kb", n, elapsed, float(np.median(lat) * 1000))
+
+
+def config3_group_fanout(seconds: float) -> None:
+    """256 agents / 8 groups of 32; group messages via the single-slot
+    fan-out kernel; priority-ordered dequeue (the priority-sort kernel)."""
+    from swarmdb_amd.runtime.engine import VIS_GROUP
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    n_agents, n_groups = 256, 8
+    cfg = QueueConfig(use_gpu=True, auto_save=False, max_agents=256,
+                      num_slots=1 << 20, staging_batch=16384,
+                      inbox_capacity=1 << 16)
+    eng = GpuEngine(cfg)
+    for a in range(n_agents):
+        eng.register_agent(a)
+    rng = np.random.default_rng(0)
+    group_bitmaps = []
+    for gi in range(n_groups):
+        bits = np.zeros(cfg.max_agents, dtype=bool)
+        bits[gi * 32 : (gi + 1) * 32] = True
+        group_bitmaps.append(eng.alloc_bitmap(bits))
+    batch = 4096  # group messages per tick; each fans out to 32 members
+    stride = 1024
+    recs = np.zeros(batch, dtype=REC_DTYPE)
+    gidx = rng.integers(0, n_groups, batch)
+    recs["sender"] = (gidx * 32).astype(np.uint32)  # a member of the group
+    recs["receiver"] = 0xFFFFFFFF  # BROADCAST routing, group-filtered
+    recs["priority"] = rng.integers(0, 4, batch)
+    recs["vis_mode"] = VIS_GROUP
+    recs["bitmap"] = np.array(group_bitmaps, dtype=np.uint32)[gidx]
+    recs["payload_len"] = stride
+    recs["content_len"] = stride
+    recs["payload_off"] = np.arange(batch, dtype=np.uint64) * stride
+    payload = rng.integers(32, 127, batch * stride, dtype=np.uint8).tobytes()
+    agents = np.arange(n_agents, dtype=np.uint32)
+    K = batch * 32 // n_agents * 2
+    lat = []
+    delivered = 0
+    for _ in range(3):
+        eng.enqueue_batch(recs, payload)
+        eng.receive_many(agents, K, priority_order=True)
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        eng.enqueue_batch(recs, payload)
+        counts, seqs = eng.receive_many(agents, K, priority_order=True)
+        eng.deliver_payloads(seqs, stride)
+        lat.append(time.perf_counter() - s)
+        delivered += int(counts.sum())
+    elapsed = time.perf_counter() - t0
+    eng.close()
+    _emit(3, "group-fanout-priority", delivered, elapsed,
+          float(np.median(lat) * 1000),
+          {"groups": n_groups, "agents": n_agents,
+           "fanout_per_group_msg": 32})
+
+
+def config5_loadbalancer(seconds: float) -> None:
+    """8 mock backends (1 per GPU on a full node); least-loaded dispatch
+    via the wavefront min-reduce kernel at >=100k req/s, with concurrent
+    JSON history spill of live queue traffic."""
+    use_gpu = _gpu_available()
+    cfg = QueueConfig(use_gpu=use_gpu, auto_save=False, max_agents=1024,
+                      num_slots=1 << 18, staging_batch=16384)
+    db = SwarmsDB(config=cfg)
+    for i in range(8):
+        db.register_llm_backend(f"backend{i}")
+    db.set_llm_load_balancing(True)
+    # background queue traffic so the spill has something to write
+    rng = np.random.default_rng(0)
+    senders = np.arange(0, 64)
+    for a in senders:
+        db.engine.register_agent(int(a))
+    recs, payload = _make_batch(rng, 4096, senders, senders, 256)
+    db.engine.enqueue_batch(recs, payload)
+
+    dispatch_batch = 8192
+    lat = []
+    n = 0
+    spills = 0
+    t0 = time.perf_counter()
+    next_spill = t0 + 1.0
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        choices = db.engine.dispatch_batch(dispatch_batch, 8)
+        lat.append(time.perf_counter() - s)
+        n += len(choices)
+        # completions keep loads bounded
+        loads = db.engine.backend_loads()[:8]
+        for b in range(8):
+            if loads[b] > 0:
+                db.engine.backend_add_load(b, -int(loads[b] * 3 // 4))
+        if time.perf_counter() >= next_spill:
+            db.save_message_history()  # JSON history spill
+            spills += 1
+            next_spill += 1.0
+    elapsed = time.perf_counter() - t0
+    loads = db.engine.backend_loads()[:8]
+    db.config.auto_save = False
+    db.close()
+    _emit(5, "llm-loadbalancer", n, elapsed,
+          float(np.median(lat) * 1000 / dispatch_batch),
+          {"unit": "requests", "backends": 8, "spills": spills,
+           "final_load_spread": int(loads.max() - loads.min()),
+           "engine": "gpu" if use_gpu else "cpu"})
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--config", type=int, required=True,
+                    choices=[1, 2, 3, 4, 5])
+    ap.add_argument("--seconds", type=float, default=5.0)
+    args = ap.parse_args()
+    if args.config == 1:
+        config1_cpu_loopback(args.seconds)
+    elif args.config == 2:
+        config2_p2p_gpu(args.seconds)
+    elif args.config == 3:
+        config3_group_fanout(args.seconds)
+    elif args.config == 4:
+        # config 4 IS the flagship bench with 4096 global agents
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        sys.argv = [
+            "bench.py",
+            "--agents", str(max(1, 4096 // world)),
+            "--steps", "30",
+            "--warmup", "5",
+        ]
+        import bench
+
+        return bench.main()
+    elif args.config == 5:
+        config5_loadbalancer(args.seconds)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

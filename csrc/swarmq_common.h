@@ -33,7 +33,9 @@ constexpr int N_STATUS = 6;
 constexpr int N_TYPES = 7;
 
 constexpr u8 VIS_ALL = 0;
-constexpr u8 VIS_BITMAP = 1;
+constexpr u8 VIS_BITMAP = 1; // filtered at dequeue (broadcast semantics)
+constexpr u8 VIS_GROUP = 2;  // filtered at fan-out (group semantics: only
+                             // member inboxes get the entry)
 
 // ---- message record: 48 B, == numpy REC_DTYPE bit-for-bit ----
 struct __attribute__((aligned(16))) Rec {

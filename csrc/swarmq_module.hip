@@ -45,6 +45,30 @@ using namespace swarmq;
 
 using ull = unsigned long long;
 
+// multi-threaded host memcpy for large staging copies (a single-threaded
+// 16 MB memcpy at ~12 GB/s would cost more than the H2D DMA it feeds)
+#include <thread>
+static void par_memcpy(void *dst, const void *src, size_t n) {
+  constexpr size_t kCut = 2u << 20;
+  if (n < kCut) {
+    std::memcpy(dst, src, n);
+    return;
+  }
+  const int nt = 4;
+  std::thread ts[nt];
+  const size_t chunk = (n + nt - 1) / nt;
+  for (int t = 0; t < nt; ++t) {
+    const size_t off = (size_t)t * chunk;
+    const size_t len = off < n ? std::min(chunk, n - off) : 0;
+    ts[t] = std::thread([=] {
+      if (len)
+        std::memcpy((char *)dst + off, (const char *)src + off, len);
+    });
+  }
+  for (auto &t : ts)
+    t.join();
+}
+
 // ---------------------------------------------------------------------------
 // kernels
 // ---------------------------------------------------------------------------
@@ -65,41 +89,57 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
                           ull *__restrict__ by_status, ull *__restrict__ sent,
                           u64 *__restrict__ bcast_list,
                           u32 *__restrict__ bcast_count, QueueGeom g) {
+  // per-block type histogram: one global atomic per bin per block instead
+  // of one per message (single hot counter words otherwise serialize the
+  // whole batch through L2 atomics)
+  __shared__ u32 h_type[N_TYPES];
+  __shared__ u32 h_msgs;
+  if (threadIdx.x < N_TYPES)
+    h_type[threadIdx.x] = 0;
+  if (threadIdx.x == N_TYPES)
+    h_msgs = 0;
+  __syncthreads();
+
   const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
-  if (wave >= n)
-    return;
-  const Rec r = stage[wave];
-  const u64 seq = base_seq + (u64)wave;
-  const u32 slot = (u32)(seq % g.num_slots);
+  if (wave < n) {
+    const Rec r = stage[wave];
+    const u64 seq = base_seq + (u64)wave;
+    const u32 slot = (u32)(seq % g.num_slots);
 
-  // payload copy, 16 B per lane per round (staging offsets are 16-B
-  // aligned; slots are slot_bytes-strided so destination is aligned too)
-  const uint4 *src =
-      reinterpret_cast<const uint4 *>(stage_pay + r.payload_off);
-  uint4 *dst =
-      reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
-  const u32 nchunk = (r.payload_len + 15u) >> 4;
-  for (u32 c = lane; c < nchunk; c += 64)
-    dst[c] = src[c];
+    // payload copy, 16 B per lane per round (staging offsets are 16-B
+    // aligned; slots are slot_bytes-strided so destination is aligned too)
+    const uint4 *src =
+        reinterpret_cast<const uint4 *>(stage_pay + r.payload_off);
+    uint4 *dst =
+        reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
+    const u32 nchunk = (r.payload_len + 15u) >> 4;
+    for (u32 c = lane; c < nchunk; c += 64)
+      dst[c] = src[c];
 
-  if (lane == 0) {
-    Rec h = r;
-    h.payload_off = (u64)slot * g.slot_bytes;
-    hdr[slot] = h;
-    status[slot] = ST_DELIVERED;
-    atomicAdd(&by_type[r.type], 1ull);
-    atomicAdd(&by_status[ST_DELIVERED], 1ull);
-    atomicAdd(&sent[r.sender], 1ull);
-    if (r.receiver == BROADCAST) {
-      u32 bi = atomicAdd(bcast_count, 1u);
-      bcast_list[bi] = seq;
-    } else {
-      ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
-      inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
-          seq;
+    if (lane == 0) {
+      Rec h = r;
+      h.payload_off = (u64)slot * g.slot_bytes;
+      hdr[slot] = h;
+      status[slot] = ST_DELIVERED;
+      atomicAdd(&h_type[r.type], 1u);
+      atomicAdd(&h_msgs, 1u);
+      atomicAdd(&sent[r.sender], 1ull);
+      if (r.receiver == BROADCAST) {
+        u32 bi = atomicAdd(bcast_count, 1u);
+        bcast_list[bi] = seq;
+      } else {
+        ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
+        inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
+            seq;
+      }
     }
   }
+  __syncthreads();
+  if (threadIdx.x < N_TYPES && h_type[threadIdx.x])
+    atomicAdd(&by_type[threadIdx.x], (ull)h_type[threadIdx.x]);
+  if (threadIdx.x == N_TYPES && h_msgs)
+    atomicAdd(&by_status[ST_DELIVERED], (ull)h_msgs);
 }
 
 // Broadcast fan-out: one thread per agent appends the batch's broadcast
@@ -109,6 +149,8 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
 __global__ void k_fanout(const u64 *__restrict__ bcast_list,
                          const u32 *__restrict__ bcast_count,
                          const u32 *__restrict__ active,
+                         const Rec *__restrict__ hdr,
+                         const u64 *__restrict__ bitmaps,
                          u64 *__restrict__ inbox, ull *__restrict__ inbox_wpos,
                          QueueGeom g) {
   const u32 a = blockIdx.x * blockDim.x + threadIdx.x;
@@ -117,11 +159,22 @@ __global__ void k_fanout(const u64 *__restrict__ bcast_list,
   const u32 nb = *bcast_count;
   if (nb == 0)
     return;
-  const ull base = inbox_wpos[a];
+  ull pos = inbox_wpos[a];
   u64 *ib = inbox + (u64)a * g.inbox_capacity;
-  for (u32 i = 0; i < nb; ++i)
-    ib[(base + i) % g.inbox_capacity] = bcast_list[i];
-  inbox_wpos[a] = base + nb;
+  for (u32 i = 0; i < nb; ++i) {
+    const u64 seq = bcast_list[i];
+    const Rec &h = hdr[seq % g.num_slots]; // uniform per i, L2-cached
+    if (h.vis_mode == VIS_GROUP) {
+      // group fan-out: only member inboxes get the entry
+      if (h.bitmap == NO_BITMAP ||
+          !((bitmaps[(u64)h.bitmap * g.bitmap_words + (a >> 6)] >>
+             (a & 63)) & 1ull))
+        continue;
+    }
+    ib[pos % g.inbox_capacity] = seq;
+    ++pos;
+  }
+  inbox_wpos[a] = pos;
 }
 
 // Dequeue: one 256-thread workgroup per polling agent. Drains the
@@ -168,7 +221,8 @@ __global__ void __launch_bounds__(256)
       if (status[slot] != ST_DELETED) {
         const Rec h = hdr[slot];
         bool vis = true;
-        if (h.vis_mode == VIS_BITMAP && h.bitmap != NO_BITMAP) {
+        if ((h.vis_mode == VIS_BITMAP || h.vis_mode == VIS_GROUP) &&
+            h.bitmap != NO_BITMAP) {
           const u64 wbits =
               bitmaps[(u64)h.bitmap * g.bitmap_words + (a >> 6)];
           vis = (wbits >> (a & 63)) & 1ull;
@@ -261,7 +315,7 @@ __global__ void k_gather(const u64 *__restrict__ seqs, int n,
                          Rec *__restrict__ out_hdr,
                          u32 *__restrict__ out_status,
                          u8 *__restrict__ out_pay, u64 evict_base,
-                         QueueGeom g) {
+                         u32 out_stride, QueueGeom g) {
   const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   if (wave >= n)
@@ -279,8 +333,11 @@ __global__ void k_gather(const u64 *__restrict__ seqs, int n,
   const Rec h = hdr[slot];
   const uint4 *src = reinterpret_cast<const uint4 *>(payload + h.payload_off);
   uint4 *dst =
-      reinterpret_cast<uint4 *>(out_pay + (u64)wave * g.slot_bytes);
-  const u32 nchunk = (h.payload_len + 15u) >> 4;
+      reinterpret_cast<uint4 *>(out_pay + (u64)wave * out_stride);
+  u32 plen = h.payload_len;
+  if (plen > out_stride)
+    plen = out_stride;
+  const u32 nchunk = (plen + 15u) >> 4;
   for (u32 c = lane; c < nchunk; c += 64)
     dst[c] = src[c];
   if (lane == 0) {
@@ -625,7 +682,7 @@ public:
       py::gil_scoped_release nogil;
       std::memcpy(h_recs_, ri.ptr, (size_t)n * sizeof(Rec));
       if (pay_bytes)
-        std::memcpy(h_pay_, pi.ptr, pay_bytes);
+        par_memcpy(h_pay_, pi.ptr, pay_bytes);
       HIP_CHECK(hipMemcpyAsync(d_stage_recs_, h_recs_, (size_t)n * sizeof(Rec),
                                hipMemcpyHostToDevice, stream_));
       if (pay_bytes)
@@ -640,7 +697,7 @@ public:
                          d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
                          0, stream_, d_bcast_, d_bcast_count_, d_active_,
-                         d_inbox_, d_wpos_, g_);
+                         d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
       HIP_CHECK(hipStreamSynchronize(stream_));
     }
     count_ = base + (u64)n;
@@ -664,7 +721,7 @@ public:
       py::gil_scoped_release nogil;
       std::memcpy(h_recs_, ri.ptr, (size_t)n * sizeof(Rec));
       if (pay_bytes)
-        std::memcpy(h_pay_, pi.ptr, pay_bytes);
+        par_memcpy(h_pay_, pi.ptr, pay_bytes);
       HIP_CHECK(hipMemcpyAsync(d_stage_recs_, h_recs_, (size_t)n * sizeof(Rec),
                                hipMemcpyHostToDevice, stream_));
       if (pay_bytes)
@@ -678,7 +735,7 @@ public:
                          d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
                          0, stream_, d_bcast_, d_bcast_count_, d_active_,
-                         d_inbox_, d_wpos_, g_);
+                         d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
     }
     count_ = base + (u64)n;
     if (count_ > g_.num_slots)
@@ -756,7 +813,7 @@ public:
       hipLaunchKernelGGL(k_gather, dim3((n + 3) / 4), dim3(256), 0,
                          copy_stream_, d_seqs_in_, n, d_hdr_, d_status_,
                          d_payload_, d_fetch_hdr_, d_fetch_status_,
-                         d_fetch_pay_, evict_base_, g_);
+                         d_fetch_pay_, evict_base_, g_.slot_bytes, g_);
       HIP_CHECK(hipMemcpyAsync(h_fetch_hdr_, d_fetch_hdr_, n * sizeof(Rec),
                                hipMemcpyDeviceToHost, copy_stream_));
       HIP_CHECK(hipMemcpyAsync(h_fetch_status_, d_fetch_status_,
@@ -780,12 +837,15 @@ public:
   // Gather + D2H into pinned host memory WITHOUT building Python
   // objects — the hot-path delivery step (payload bytes land in host
   // RAM; zero per-message host work). Returns payload bytes landed.
-  u64 fetch_raw(py::array_t<u64> seqs) {
+  u64 fetch_raw(py::array_t<u64> seqs, u32 stride) {
     const int n = (int)seqs.size();
     if (n == 0)
       return 0;
     if ((u32)n > staging_batch_)
       throw std::invalid_argument("fetch batch exceeds staging_batch");
+    if (stride == 0 || stride > g_.slot_bytes)
+      stride = g_.slot_bytes;
+    stride = (stride + 15u) & ~15u;
     u64 bytes = 0;
     {
       py::gil_scoped_release nogil;
@@ -796,11 +856,11 @@ public:
       hipLaunchKernelGGL(k_gather, dim3((n + 3) / 4), dim3(256), 0,
                          copy_stream_, d_seqs_in_, n, d_hdr_, d_status_,
                          d_payload_, d_fetch_hdr_, d_fetch_status_,
-                         d_fetch_pay_, evict_base_, g_);
+                         d_fetch_pay_, evict_base_, stride, g_);
       HIP_CHECK(hipMemcpyAsync(h_fetch_hdr_, d_fetch_hdr_, n * sizeof(Rec),
                                hipMemcpyDeviceToHost, copy_stream_));
       HIP_CHECK(hipMemcpyAsync(h_fetch_pay_, d_fetch_pay_,
-                               (size_t)n * g_.slot_bytes,
+                               (size_t)n * stride,
                                hipMemcpyDeviceToHost, copy_stream_));
       HIP_CHECK(hipStreamSynchronize(copy_stream_));
       for (int i = 0; i < n; ++i)
@@ -1109,7 +1169,7 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
       .def("receive_many", &DeviceQueue::receive_many)
       .def("fetch", &DeviceQueue::fetch)
-      .def("fetch_raw", &DeviceQueue::fetch_raw)
+      .def("fetch_raw", &DeviceQueue::fetch_raw, py::arg("seqs"), py::arg("stride") = 0)
       .def("set_status", &DeviceQueue::set_status)
       .def("get_status", &DeviceQueue::get_status)
       .def("query_range", &DeviceQueue::query_range)

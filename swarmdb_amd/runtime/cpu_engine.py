@@ -33,6 +33,7 @@ from .engine import (
     ST_READ,
     VIS_ALL,
     VIS_BITMAP,
+    VIS_GROUP,
     Engine,
 )
 
@@ -161,8 +162,23 @@ class CpuEngine(Engine):
             bmask = recv == BROADCAST
             if bmask.any():
                 bseqs = seqs[bmask]
-                for a in np.flatnonzero(self._active):
-                    self._inbox[int(a)].append_many(bseqs)
+                gmask = recs["vis_mode"][bmask] == VIS_GROUP
+                plain = bseqs[~gmask]
+                if len(plain):
+                    for a in np.flatnonzero(self._active):
+                        self._inbox[int(a)].append_many(plain)
+                # group fan-out: member inboxes only
+                for s, bm in zip(bseqs[gmask], recs["bitmap"][bmask][gmask]):
+                    members = np.flatnonzero(
+                        self._bitmaps[int(bm)] & self._active
+                    )
+                    one = np.array([s], dtype=np.uint64)
+                    for a in members:
+                        ai = int(a)
+                        if ai not in self._inbox:
+                            self._inbox[ai] = _U64Ring()
+                            self._recv_ts[ai] = _U64Ring()
+                        self._inbox[ai].append_many(one)
             pmask = ~bmask
             if pmask.any():
                 prs = recv[pmask]
@@ -195,7 +211,7 @@ class CpuEngine(Engine):
         st = self._status[seqs]
         ok = st != ST_DELETED
         vm = self._hdr["vis_mode"][seqs]
-        restricted = vm == VIS_BITMAP
+        restricted = (vm == VIS_BITMAP) | (vm == VIS_GROUP)
         if restricted.any():
             bidx = self._hdr["bitmap"][seqs[restricted]]
             allowed = np.fromiter(

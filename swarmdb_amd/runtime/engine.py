@@ -59,7 +59,11 @@ TYPE_CODES = {n: i for i, n in enumerate(TYPE_NAMES)}
 
 # visibility modes
 VIS_ALL = 0      # no restriction (visible_to empty)
-VIS_BITMAP = 1   # restricted: agent must be set in the bitmap
+VIS_BITMAP = 1   # restricted: agent must be set in the bitmap (inbox entry
+#                  still lands in every inbox, filtered at dequeue — the
+#                  reference's broadcast-listing behavior)
+VIS_GROUP = 2    # group fan-out: inbox entries land ONLY in member inboxes
+#                  (one slot per group message; the fan-out kernel filters)
 
 # record flags
 FLAG_HAS_EXTRAS = 1   # payload tail carries an extras-JSON blob

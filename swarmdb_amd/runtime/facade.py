@@ -64,6 +64,7 @@ from .engine import (
     TYPE_NAMES,
     VIS_ALL,
     VIS_BITMAP,
+    VIS_GROUP,
     Engine,
 )
 
@@ -368,6 +369,72 @@ class SwarmsDB:
                     )
                 )
             return ids
+
+    def send_to_group_fast(
+        self,
+        group_name: str,
+        sender_id: str,
+        content: Union[str, Dict[str, Any], List[Any]],
+        message_type: Union[MessageType, str] = MessageType.CHAT,
+        priority: Union[MessagePriority, int] = MessagePriority.NORMAL,
+        metadata: Optional[Dict[str, Any]] = None,
+    ) -> str:
+        """Single-slot group fan-out (the GPU-native group path,
+        BASELINE config 3): ONE message slot, the fan-out kernel appends
+        the entry to member inboxes only (VIS_GROUP). Returns one message
+        id — use :meth:`send_to_group` for the reference-compatible
+        one-id-per-member behavior."""
+        with self._lock:
+            groups = self.metadata.get("agent_groups", {})
+            if group_name not in groups:
+                raise ValueError(f"group '{group_name}' does not exist")
+            self.register_agent(sender_id)
+            members = [a for a in groups[group_name] if a != sender_id]
+            md = dict(metadata or {})
+            md["group"] = group_name
+            mtype = MessageType(message_type)
+            prio = (
+                priority
+                if isinstance(priority, MessagePriority)
+                else MessagePriority(priority)
+            )
+            msg = Message(
+                sender_id=sender_id,
+                receiver_id=None,
+                content=content,
+                type=mtype,
+                priority=prio,
+                metadata=md,
+                token_count=self._count_tokens(content),
+                visible_to=members,
+            )
+            rec = np.zeros(1, dtype=REC_DTYPE)
+            content_b, is_json = encode_content(content)
+            extras_b = encode_extras(msg.id, md, members)
+            payload = content_b + extras_b
+            rec["sender"] = self._agent_idx[sender_id]
+            rec["receiver"] = BROADCAST
+            rec["type"] = _type_code(mtype)
+            rec["priority"] = prio.value
+            rec["timestamp"] = msg.timestamp
+            rec["token_count"] = msg.token_count or 0
+            rec["payload_len"] = len(payload)
+            rec["content_len"] = len(content_b)
+            rec["flags"] = FLAG_HAS_EXTRAS | (FLAG_JSON_CONTENT if is_json else 0)
+            rec["vis_mode"] = VIS_GROUP
+            rec["bitmap"] = self._group_bitmap(group_name, members)
+            seqs = self.engine.enqueue_batch(rec, payload)
+            self._id_to_seq[msg.id] = int(seqs[0])
+            self._maybe_autosave()
+            return msg.id
+
+    def _group_bitmap(self, group_name: str, members: List[str]) -> int:
+        """Cached per-(group, membership) visibility bitmap."""
+        key = (group_name, tuple(members))
+        cache = self.metadata.setdefault("_group_bitmaps", {})
+        if key not in cache:
+            cache[key] = self._bitmap_for(members)
+        return cache[key]
 
     def resend_failed_messages(self) -> List[str]:
         """Re-send every FAILED message as a NEW message, linked via

@@ -220,17 +220,20 @@ class GpuEngine(Engine):
             done += chunk
         return out, pays
 
-    def deliver_payloads(self, seqs: np.ndarray) -> int:
+    def deliver_payloads(self, seqs: np.ndarray, max_payload: int = 0) -> int:
         """Gather + D2H the payloads of `seqs` into pinned host memory
         (the delivery step of the hot path) without building per-message
-        Python objects. Returns bytes landed."""
+        Python objects. `max_payload` (if known) tightens the D2H stride
+        below slot_bytes. Returns bytes landed."""
         seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
         n = len(seqs)
         done = 0
         total = 0
         while done < n:
             chunk = min(self._staging, n - done)
-            total += int(self.q.fetch_raw(seqs[done : done + chunk]))
+            total += int(
+                self.q.fetch_raw(seqs[done : done + chunk], int(max_payload))
+            )
             done += chunk
         return total
 

@@ -352,3 +352,24 @@ def test_resend_failed(tmp_db, monkeypatch):
     assert "resent_from" in m.metadata
     # second resend is a no-op
     assert tmp_db.resend_failed_messages() == []
+
+
+def test_send_to_group_fast(tmp_db):
+    tmp_db.add_agent_group("team", ["a", "b", "c", "d"])
+    mid = tmp_db.send_to_group_fast("team", "a", "one slot fanout",
+                                    priority=MessagePriority.HIGH)
+    # one message id; delivered to members only, not the sender
+    for member in ["b", "c", "d"]:
+        got = tmp_db.receive_messages(member, timeout=0)
+        assert [m.id for m in got] == [mid]
+        assert got[0].metadata["group"] == "team"
+        assert got[0].priority == MessagePriority.HIGH
+    assert tmp_db.receive_messages("a", timeout=0) == []
+    # non-members never see it in their inbox (group fan-out filters at
+    # append, unlike broadcast)
+    tmp_db.register_agent("outsider")
+    assert tmp_db.get_agent_messages("outsider") == []
+    # member inboxes list it
+    assert [m.id for m in tmp_db.get_agent_messages("b")] == [mid]
+    with pytest.raises(ValueError):
+        tmp_db.send_to_group_fast("nope", "a", "x")
