@@ -462,6 +462,60 @@ def create_app(
             total_messages=total,
         )
 
+    @app.get("/metrics")
+    async def metrics():
+        """Prometheus exposition of the engine's running counters (the
+        reference had pull-only O(N) stats scans and no metrics endpoint —
+        SURVEY.md §5.5). No auth, like /health."""
+        from prometheus_client import (
+            CONTENT_TYPE_LATEST,
+            CollectorRegistry,
+            Gauge,
+            generate_latest,
+        )
+        from starlette.responses import Response
+
+        from ..runtime.engine import STATUS_NAMES, TYPE_NAMES
+
+        reg = CollectorRegistry()
+        stats = db.engine.stats_arrays()
+        g_total = Gauge("swarmdb_messages_total", "Total messages enqueued",
+                        registry=reg)
+        g_total.set(db.engine.total_messages())
+        g_agents = Gauge("swarmdb_registered_agents", "Registered agents",
+                         registry=reg)
+        g_agents.set(len(db.registered_agents))
+        g_type = Gauge("swarmdb_messages_by_type", "Messages by type",
+                       ["type"], registry=reg)
+        for i, name in enumerate(TYPE_NAMES):
+            g_type.labels(type=name).set(int(stats["by_type"][i]))
+        g_status = Gauge("swarmdb_messages_by_status", "Messages by status",
+                         ["status"], registry=reg)
+        for i, name in enumerate(STATUS_NAMES):
+            g_status.labels(status=name).set(int(stats["by_status"][i]))
+        g_load = Gauge("swarmdb_backend_load", "LLM backend in-flight load",
+                       ["backend"], registry=reg)
+        loads = db.engine.backend_loads()
+        for bid, idx in db._llm_backend_idx.items():
+            g_load.labels(backend=bid).set(int(loads[idx]))
+        return Response(generate_latest(reg), media_type=CONTENT_TYPE_LATEST)
+
+    @app.get("/admin/trace")
+    async def trace_summary(
+        enable: Optional[bool] = Query(None),
+        current: str = Depends(get_current_agent),
+    ):
+        """Tracer control + per-op latency summary (new; the reference has
+        no tracing)."""
+        _require_admin(current)
+        from ..utils.tracing import tracer
+
+        if enable is True:
+            tracer.enable()
+        elif enable is False:
+            tracer.disable()
+        return {"enabled": tracer.enabled, "ops": tracer.summary()}
+
     @app.get("/stats", response_model=SystemStats)
     async def get_stats(current: str = Depends(get_current_agent)):
         """reference api.py:818-838 (admin only)."""

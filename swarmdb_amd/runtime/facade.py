@@ -46,6 +46,7 @@ from ..core.wire import (
     parse_derived_id,
 )
 from ..utils.hashing import partition_for
+from ..utils.tracing import tracer
 from .engine import (
     BROADCAST,
     FLAG_DERIVED_ID,
@@ -470,7 +471,8 @@ class SwarmsDB:
         """Zero-per-message-Python send: REC_DTYPE array + one payload
         buffer straight into the engine (pinned staging + one kernel on
         GPU). Message ids are derived from (rank, seq). Returns seqs."""
-        seqs = self.engine.enqueue_batch(recs, payloads)
+        with tracer.span("send_batch", n=len(recs)):
+            seqs = self.engine.enqueue_batch(recs, payloads)
         self._maybe_autosave()
         return seqs
 
@@ -480,7 +482,10 @@ class SwarmsDB:
     ):
         """Drain many agents in one engine call (one dequeue kernel on
         GPU). Returns (counts, seqs)."""
-        return self.engine.receive_many(agent_idxs, max_per_agent, priority_order)
+        with tracer.span("receive_batch", n=len(agent_idxs)):
+            return self.engine.receive_many(
+                agent_idxs, max_per_agent, priority_order
+            )
 
     def agent_index(self, agent_id: str) -> int:
         with self._lock:

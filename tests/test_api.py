@@ -316,3 +316,39 @@ def test_rate_limiter():
     assert all(rl.allow("1.2.3.4") for _ in range(3))
     assert not rl.allow("1.2.3.4")
     assert rl.allow("5.6.7.8")  # other IPs unaffected
+
+
+def test_metrics_endpoint(client):
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "x"})
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert "swarmdb_messages_total 1.0" in body
+    assert 'swarmdb_messages_by_status{status="delivered"} 1.0' in body
+    assert "swarmdb_registered_agents 2.0" in body
+
+
+def test_trace_endpoint(client):
+    ha = auth(client, "admin")
+    assert client.get("/admin/trace", headers=auth(client, "x")).status_code == 403
+    r = client.get("/admin/trace?enable=true", headers=ha)
+    assert r.json()["enabled"] is True
+    import numpy as np
+
+    from swarmdb_amd.runtime.engine import NO_BITMAP, REC_DTYPE, VIS_ALL
+
+    a = client.db.agent_index("a")
+    recs = np.zeros(4, dtype=REC_DTYPE)
+    recs["sender"] = a
+    recs["receiver"] = a
+    recs["payload_len"] = 8
+    recs["content_len"] = 8
+    recs["payload_off"] = np.arange(4, dtype=np.uint64) * 8
+    recs["vis_mode"] = VIS_ALL
+    recs["bitmap"] = NO_BITMAP
+    client.db.send_batch(recs, b"x" * 32)
+    r = client.get("/admin/trace", headers=ha)
+    assert "send_batch" in r.json()["ops"]
+    r = client.get("/admin/trace?enable=false", headers=ha)
+    assert r.json()["enabled"] is False
