@@ -577,18 +577,24 @@ public:
     HIP_CHECK(hipMalloc(&d_out_seqs_, out_pool_ * sizeof(u64)));
     HIP_CHECK(hipMalloc(&d_out_counts_, (size_t)max_agents * sizeof(u32)));
 
-    // staging (device side of the H2D batch)
-    HIP_CHECK(hipMalloc(&d_stage_recs_, (size_t)staging_batch * sizeof(Rec)));
+    // staging (device side of the H2D batch), double-buffered so the
+    // host can fill batch i+1 while batch i's H2D/kernels run
     stage_pay_bytes_ = (size_t)staging_batch * 1024; // grows on demand
-    HIP_CHECK(hipMalloc(&d_stage_pay_, stage_pay_bytes_));
+    for (int s = 0; s < 2; ++s) {
+      HIP_CHECK(hipMalloc(&d_stage_recs_[s], (size_t)staging_batch * sizeof(Rec)));
+      HIP_CHECK(hipMalloc(&d_stage_pay_[s], stage_pay_bytes_));
+      HIP_CHECK(hipEventCreateWithFlags(&stage_ev_[s], hipEventDisableTiming));
+    }
     HIP_CHECK(hipMalloc(&d_seqs_in_, (size_t)staging_batch * sizeof(u64)));
     HIP_CHECK(hipMalloc(&d_fetch_hdr_, (size_t)staging_batch * sizeof(Rec)));
     HIP_CHECK(hipMalloc(&d_fetch_status_, (size_t)staging_batch * sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_fetch_pay_, (size_t)staging_batch * slot_bytes));
 
     // pinned host staging
-    HIP_CHECK(hipHostMalloc(&h_recs_, (size_t)staging_batch * sizeof(Rec)));
-    HIP_CHECK(hipHostMalloc(&h_pay_, stage_pay_bytes_));
+    for (int s = 0; s < 2; ++s) {
+      HIP_CHECK(hipHostMalloc(&h_recs_[s], (size_t)staging_batch * sizeof(Rec)));
+      HIP_CHECK(hipHostMalloc(&h_pay_[s], stage_pay_bytes_));
+    }
     HIP_CHECK(hipHostMalloc(&h_out_seqs_, out_pool_ * sizeof(u64)));
     HIP_CHECK(hipHostMalloc(&h_out_counts_, (size_t)max_agents * sizeof(u32)));
     HIP_CHECK(hipHostMalloc(&h_fetch_hdr_, (size_t)staging_batch * sizeof(Rec)));
@@ -625,15 +631,19 @@ public:
                              d_received_, d_bcast_, d_bcast_count_,
                              d_backend_loads_, d_choices_, d_match_,
                              d_match_count_, d_needle_, d_agents_, d_unread_,
-                             d_out_seqs_, d_out_counts_, d_stage_recs_,
-                             d_stage_pay_, d_seqs_in_, d_fetch_hdr_,
+                             d_out_seqs_, d_out_counts_, d_stage_recs_[0],
+                             d_stage_recs_[1], d_stage_pay_[0],
+                             d_stage_pay_[1], d_seqs_in_, d_fetch_hdr_,
                              d_fetch_status_, d_fetch_pay_})
       (void)hipFree(p);
-    for (void *p : std::vector<void *>{h_recs_, h_pay_, h_out_seqs_,
+    for (void *p : std::vector<void *>{h_recs_[0], h_recs_[1], h_pay_[0],
+                                       h_pay_[1], h_out_seqs_,
                                        h_out_counts_, h_fetch_hdr_,
                                        h_fetch_status_, h_fetch_pay_,
                                        h_choices_})
       (void)hipHostFree(p);
+    (void)hipEventDestroy(stage_ev_[0]);
+    (void)hipEventDestroy(stage_ev_[1]);
     (void)hipEventDestroy(ev_);
     (void)hipStreamDestroy(stream_);
     (void)hipStreamDestroy(copy_stream_);
@@ -680,19 +690,21 @@ public:
     const u64 base = count_;
     {
       py::gil_scoped_release nogil;
-      std::memcpy(h_recs_, ri.ptr, (size_t)n * sizeof(Rec));
+      std::memcpy(h_recs_[0], ri.ptr, (size_t)n * sizeof(Rec));
       if (pay_bytes)
-        par_memcpy(h_pay_, pi.ptr, pay_bytes);
-      HIP_CHECK(hipMemcpyAsync(d_stage_recs_, h_recs_, (size_t)n * sizeof(Rec),
+        par_memcpy(h_pay_[0], pi.ptr, pay_bytes);
+      HIP_CHECK(hipMemcpyAsync(d_stage_recs_[0], h_recs_[0],
+                               (size_t)n * sizeof(Rec),
                                hipMemcpyHostToDevice, stream_));
       if (pay_bytes)
-        HIP_CHECK(hipMemcpyAsync(d_stage_pay_, h_pay_, pay_bytes,
+        HIP_CHECK(hipMemcpyAsync(d_stage_pay_[0], h_pay_[0], pay_bytes,
                                  hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipEventRecord(stage_ev_[0], stream_));
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
       const int waves_per_block = 4; // 256 threads
       const int blocks = (n + waves_per_block - 1) / waves_per_block;
       hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
-                         d_stage_recs_, d_stage_pay_, n, base, d_hdr_,
+                         d_stage_recs_[0], d_stage_pay_[0], n, base, d_hdr_,
                          d_status_, d_payload_, d_inbox_, d_wpos_, d_by_type_,
                          d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
@@ -706,46 +718,79 @@ public:
     return base;
   }
 
-  // Async variant for graph-friendly benchmarking: no sync; caller must
-  // call sync() before reading results.
+  // Async variant: fill slot 0 and launch without a sync; caller must
+  // sync() (or rely on stream ordering) before reading results.
   u64 enqueue_batch_async(py::buffer recs, py::buffer pay, int n) {
-    py::buffer_info ri = recs.request(), pi = pay.request();
     if (n <= 0)
       return count_;
-    if ((u32)n > staging_batch_)
-      throw std::invalid_argument("batch exceeds staging_batch");
-    const size_t pay_bytes = (size_t)pi.size * pi.itemsize;
-    ensure_stage_pay(pay_bytes + 16);
-    const u64 base = count_;
-    {
-      py::gil_scoped_release nogil;
-      std::memcpy(h_recs_, ri.ptr, (size_t)n * sizeof(Rec));
-      if (pay_bytes)
-        par_memcpy(h_pay_, pi.ptr, pay_bytes);
-      HIP_CHECK(hipMemcpyAsync(d_stage_recs_, h_recs_, (size_t)n * sizeof(Rec),
-                               hipMemcpyHostToDevice, stream_));
-      if (pay_bytes)
-        HIP_CHECK(hipMemcpyAsync(d_stage_pay_, h_pay_, pay_bytes,
-                                 hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      const int blocks = (n + 3) / 4;
-      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
-                         d_stage_recs_, d_stage_pay_, n, base, d_hdr_,
-                         d_status_, d_payload_, d_inbox_, d_wpos_, d_by_type_,
-                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
-      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
-                         0, stream_, d_bcast_, d_bcast_count_, d_active_,
-                         d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
-    }
-    count_ = base + (u64)n;
-    if (count_ > g_.num_slots)
-      evict_base_ = count_ - g_.num_slots;
-    return base;
+    stage_fill(0, recs, pay, n);
+    return enqueue_staged(0);
   }
 
   void sync() {
     py::gil_scoped_release nogil;
     HIP_CHECK(hipStreamSynchronize(stream_));
+  }
+
+  // ---- pipelined staging: fill one pinned set while the other's
+  // H2D/kernels are in flight ----
+
+  void stage_fill(int slot, py::buffer recs, py::buffer pay, int n) {
+    if (slot < 0 || slot > 1)
+      throw std::invalid_argument("slot must be 0 or 1");
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("batch exceeds staging_batch");
+    py::buffer_info ri = recs.request(), pi = pay.request();
+    const size_t pay_bytes = (size_t)pi.size * pi.itemsize;
+    ensure_stage_pay(pay_bytes + 16);
+    {
+      py::gil_scoped_release nogil;
+      // don't overwrite pinned memory a previous H2D still reads
+      HIP_CHECK(hipEventSynchronize(stage_ev_[slot]));
+      std::memcpy(h_recs_[slot], ri.ptr, (size_t)n * sizeof(Rec));
+      if (pay_bytes)
+        par_memcpy(h_pay_[slot], pi.ptr, pay_bytes);
+    }
+    staged_n_[slot] = n;
+    staged_pay_[slot] = pay_bytes;
+  }
+
+  // Launch H2D + enqueue kernels for a previously filled slot; returns
+  // the base seq. NO sync — receive_many on the same stream is ordered
+  // after it.
+  u64 enqueue_staged(int slot) {
+    if (slot < 0 || slot > 1)
+      throw std::invalid_argument("slot must be 0 or 1");
+    const int n = staged_n_[slot];
+    if (n <= 0)
+      return count_;
+    const size_t pay_bytes = staged_pay_[slot];
+    const u64 base = count_;
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot], h_recs_[slot],
+                               (size_t)n * sizeof(Rec),
+                               hipMemcpyHostToDevice, stream_));
+      if (pay_bytes)
+        HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot], h_pay_[slot], pay_bytes,
+                                 hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipEventRecord(stage_ev_[slot], stream_));
+      HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
+      const int blocks = (n + 3) / 4;
+      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
+                         d_stage_recs_[slot], d_stage_pay_[slot], n, base,
+                         d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
+                         d_by_type_, d_by_status_, d_sent_, d_bcast_,
+                         d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
+                         dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
+                         d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
+                         g_);
+    }
+    count_ = base + (u64)n;
+    if (count_ > g_.num_slots)
+      evict_base_ = count_ - g_.num_slots;
+    return base;
   }
 
   u32 alloc_bitmap(py::buffer words) {
@@ -1077,10 +1122,12 @@ private:
     while (nb < bytes)
       nb *= 2;
     HIP_CHECK(hipStreamSynchronize(stream_));
-    HIP_CHECK(hipFree(d_stage_pay_));
-    (void)hipHostFree(h_pay_);
-    HIP_CHECK(hipMalloc(&d_stage_pay_, nb));
-    HIP_CHECK(hipHostMalloc(&h_pay_, nb));
+    for (int s = 0; s < 2; ++s) {
+      HIP_CHECK(hipFree(d_stage_pay_[s]));
+      (void)hipHostFree(h_pay_[s]);
+      HIP_CHECK(hipMalloc(&d_stage_pay_[s], nb));
+      HIP_CHECK(hipHostMalloc(&h_pay_[s], nb));
+    }
     stage_pay_bytes_ = nb;
   }
 
@@ -1122,15 +1169,18 @@ private:
   u32 *d_unread_{};
   u64 *d_out_seqs_{};
   u32 *d_out_counts_{};
-  Rec *d_stage_recs_{};
-  u8 *d_stage_pay_{};
+  Rec *d_stage_recs_[2] = {};
+  u8 *d_stage_pay_[2] = {};
+  int staged_n_[2] = {0, 0};
+  size_t staged_pay_[2] = {0, 0};
+  hipEvent_t stage_ev_[2] = {};
   u64 *d_seqs_in_{};
   Rec *d_fetch_hdr_{};
   u32 *d_fetch_status_{};
   u8 *d_fetch_pay_{};
 
-  Rec *h_recs_{};
-  u8 *h_pay_{};
+  Rec *h_recs_[2] = {};
+  u8 *h_pay_[2] = {};
   u64 *h_out_seqs_{};
   u32 *h_out_counts_{};
   Rec *h_fetch_hdr_{};
@@ -1166,6 +1216,8 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("enqueue_batch", &DeviceQueue::enqueue_batch)
       .def("enqueue_batch_async", &DeviceQueue::enqueue_batch_async)
       .def("sync", &DeviceQueue::sync)
+      .def("stage_fill", &DeviceQueue::stage_fill)
+      .def("enqueue_staged", &DeviceQueue::enqueue_staged)
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
       .def("receive_many", &DeviceQueue::receive_many)
       .def("fetch", &DeviceQueue::fetch)
