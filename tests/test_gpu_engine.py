@@ -406,3 +406,41 @@ def test_staged_pipeline_roundtrip(gpu_engine):
     hdrs, pays = eng.fetch(check)
     for s, p in zip(check, pays):
         assert p == payloads[int(s)], int(s)
+
+
+def test_group_fanout_parity():
+    """VIS_GROUP: one slot, member-only inbox entries — CPU vs GPU."""
+    from swarmdb_amd.runtime.engine import VIS_GROUP
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    cfg = small_cfg()
+    gpu = GpuEngine(cfg)
+    cpu = CpuEngine(small_cfg(use_gpu=False))
+    try:
+        rng = np.random.default_rng(5)
+        n_agents = 32
+        for a in range(n_agents):
+            gpu.register_agent(a)
+            cpu.register_agent(a)
+        bits = np.zeros(cfg.max_agents, dtype=bool)
+        bits[[2, 5, 9, 30]] = True
+        bg, bc = gpu.alloc_bitmap(bits), cpu.alloc_bitmap(bits)
+        assert bg == bc
+        recs, payload = make_batch(rng, 10, n_agents, payload_bytes=64)
+        recs["receiver"] = BROADCAST
+        recs["vis_mode"] = VIS_GROUP
+        recs["bitmap"] = bg
+        gpu.enqueue_batch(recs, payload)
+        cpu.enqueue_batch(recs, payload)
+        for a in range(n_agents):
+            pg = gpu.peek_inbox(a)
+            pc = cpu.peek_inbox(a)
+            assert (pg == pc).all(), a
+            expect = 10 if bits[a] else 0
+            assert len(pg) == expect, (a, len(pg))
+            rg = gpu.receive(a, 100)
+            rc = cpu.receive(a, 100)
+            assert (rg == rc).all(), a
+            assert len(rg) == expect
+    finally:
+        gpu.close()
