@@ -279,6 +279,64 @@ def create_app(
         msg = db.get_message(mid)
         return MessageResponse.from_message(msg)
 
+    @app.post("/messages/batch")
+    async def send_messages_batch(
+        messages: List[MessageRequest], current: str = Depends(get_current_agent)
+    ):
+        """Bulk ingestion (new — no reference analog): the whole list goes
+        through ONE engine enqueue (one pinned H2D + one kernel on GPU)."""
+        import numpy as np
+
+        from ..core.wire import derived_id, encode_content
+        from ..runtime.engine import (
+            BROADCAST,
+            NO_BITMAP,
+            REC_DTYPE,
+            VIS_ALL,
+        )
+
+        if not messages:
+            return {"status": "sent", "message_ids": []}
+        sidx = db.agent_index(current)
+        n = len(messages)
+        recs = np.zeros(n, dtype=REC_DTYPE)
+        chunks: List[bytes] = []
+        off = 0
+        import time as _time
+
+        now = _time.time()
+        for i, req in enumerate(messages):
+            content_b, is_json = encode_content(req.content)
+            pad = (-len(content_b)) % 16
+            chunks.append(content_b + b"\x00" * pad)
+            recs["sender"][i] = sidx
+            recs["receiver"][i] = (
+                BROADCAST
+                if req.receiver_id is None
+                else db.agent_index(req.receiver_id)
+            )
+            recs["type"][i] = list(MessageType).index(req.message_type)
+            recs["priority"][i] = req.priority.value
+            recs["timestamp"][i] = now
+            recs["vis_mode"][i] = VIS_ALL
+            recs["bitmap"][i] = NO_BITMAP
+            recs["payload_off"][i] = off
+            recs["payload_len"][i] = len(content_b)
+            recs["content_len"][i] = len(content_b)
+            from ..runtime.engine import FLAG_DERIVED_ID, FLAG_JSON_CONTENT
+
+            recs["flags"][i] = FLAG_DERIVED_ID | (
+                FLAG_JSON_CONTENT if is_json else 0
+            )
+            off += len(content_b) + pad
+        seqs = db.send_batch(recs, b"".join(chunks))
+        return {
+            "status": "sent",
+            "message_ids": [
+                derived_id(db.config.rank, int(s)) for s in seqs
+            ],
+        }
+
     @app.post("/messages/broadcast", response_model=BroadcastResponse)
     async def broadcast_message(
         req: BroadcastRequest, current: str = Depends(get_current_agent)

@@ -352,3 +352,26 @@ def test_trace_endpoint(client):
     assert "send_batch" in r.json()["ops"]
     r = client.get("/admin/trace?enable=false", headers=ha)
     assert r.json()["enabled"] is False
+
+
+def test_batch_send_endpoint(client):
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    client.post("/agents/register", headers=hb, json={"agent_id": "bob"})
+    msgs = [{"receiver_id": "bob", "content": f"bulk {i}"} for i in range(5)]
+    msgs.append({"receiver_id": None, "content": {"kind": "announce"}})
+    r = client.post("/messages/batch", headers=ha, json=msgs)
+    assert r.status_code == 200, r.text
+    ids = r.json()["message_ids"]
+    assert len(ids) == 6
+    got = client.post("/agents/receive?timeout=0&max_messages=100",
+                      headers=hb).json()
+    # 5 p2p + 1 broadcast
+    assert len(got) == 6
+    assert got[-1]["content"] == {"kind": "announce"}
+    # derived ids resolve through GET
+    r = client.get(f"/messages/{ids[0]}", headers=ha)
+    assert r.status_code == 200
+    assert r.json()["content"] == "bulk 0"
+    # empty batch
+    r = client.post("/messages/batch", headers=ha, json=[])
+    assert r.json()["message_ids"] == []
