@@ -1,0 +1,68 @@
+#!/usr/bin/env python3
+"""End-to-end demo — the reference's __main__ smoke flow (reference
+"swarmdb/ main.py":1398-1453): register 3 agents, point-to-point sends,
+broadcast, receive, group create + group send, stats, close.
+
+Runs on the CPU engine anywhere; uses the GPU engine automatically when
+an MI355X is visible.
+"""
+
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+from swarmdb_amd import MessagePriority, MessageType, QueueConfig, SwarmsDB
+
+
+def main() -> None:
+    cfg = QueueConfig.from_env(save_dir="demo_history", max_agents=64)
+    with SwarmsDB(config=cfg) as db:
+        print(f"engine: {type(db.engine).__name__}")
+
+        for agent in ["agent1", "agent2", "agent3"]:
+            db.register_agent(agent)
+        print(f"registered: {sorted(db.registered_agents)}")
+
+        mid = db.send_message(
+            "agent1",
+            "Hello agent2, please process this request",
+            receiver_id="agent2",
+            message_type=MessageType.CHAT,
+            priority=MessagePriority.HIGH,
+        )
+        print(f"p2p sent: {mid}")
+
+        db.send_message(
+            "agent2",
+            {"function": "analyze", "args": {"depth": 3}},
+            receiver_id="agent3",
+            message_type=MessageType.FUNCTION_CALL,
+        )
+
+        bid = db.broadcast_message(
+            "agent1", "System maintenance at 02:00", priority=MessagePriority.CRITICAL
+        )
+        print(f"broadcast sent: {bid}")
+
+        for agent in ["agent2", "agent3"]:
+            msgs = db.receive_messages(agent, timeout=0.1, priority_order=True)
+            print(f"{agent} received {len(msgs)}:")
+            for m in msgs:
+                print(f"  [{m.priority.name:8s}] {m.type.value}: {m.content!r}")
+
+        db.add_agent_group("analysts", ["agent1", "agent2", "agent3"])
+        gids = db.send_to_group("analysts", "agent1", "group sync at noon")
+        print(f"group send -> {len(gids)} messages")
+        fast_id = db.send_to_group_fast("analysts", "agent1", "fast group ping")
+        print(f"fast group send -> 1 slot ({fast_id})")
+
+        stats = db.get_stats()
+        print(f"stats: total={stats['total_messages']} "
+              f"by_type={stats['messages_by_type']} "
+              f"by_status={stats['messages_by_status']}")
+    print("closed (history saved)")
+
+
+if __name__ == "__main__":
+    main()

@@ -43,6 +43,8 @@ class QueueConfig:
     use_gpu: Optional[bool] = None       # None = auto-detect
 
     # --- persistence tier (reference swarmdb/ main.py:156-166, 221-230) ---
+    log_file: Optional[str] = None       # rotating file log (reference
+    #                                      loguru sink, swarmdb/ main.py:170-189)
     save_dir: str = "message_history"
     auto_save: bool = True
     save_interval: float = 300.0
@@ -66,6 +68,7 @@ class QueueConfig:
             num_partitions=int(env.get("KAFKA_NUM_PARTITIONS", "3")),
             replication_factor=int(env.get("KAFKA_REPLICATION_FACTOR", "1")),
             save_dir=env.get("MESSAGE_HISTORY_DIR", "message_history"),
+            log_file=env.get("LOG_FILE") or None,
             save_interval=float(env.get("SAVE_INTERVAL_SECONDS", "300")),
             max_agents=int(env.get("SWARMQ_MAX_AGENTS", "8192")),
             slot_bytes=int(env.get("SWARMQ_SLOT_BYTES", "2048")),

@@ -111,6 +111,20 @@ class SwarmsDB:
         if max_messages_per_file is not None:
             self.config.max_messages_per_file = max_messages_per_file
 
+        if self.config.log_file:
+            # reference loguru sink semantics: 10 MB rotation, INFO
+            # (swarmdb/ main.py:170-189), stdlib implementation
+            from logging.handlers import RotatingFileHandler
+
+            handler = RotatingFileHandler(
+                self.config.log_file, maxBytes=10 * 1024 * 1024, backupCount=5
+            )
+            handler.setFormatter(logging.Formatter(
+                "%(asctime)s | %(levelname)s | %(name)s | %(message)s"
+            ))
+            logger.addHandler(handler)
+            logger.setLevel(logging.INFO)
+
         self.token_counter = token_counter
         if engine is None:
             engine = self._default_engine(self.config)
