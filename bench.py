@@ -138,11 +138,18 @@ def main() -> int:
         engine.register_agent(int(a))
 
     router = None
+    grouter = None
     if dist_on:
-        from swarmdb_amd.parallel.router import CrossGpuRouter
+        if have_gpu:
+            # GPU-direct: pack kernel -> RCCL all-to-all over xGMI ->
+            # device-side ingest (payloads never touch the host)
+            from swarmdb_amd.parallel.router import GpuDirectRouter
 
-        device = torch.device("cuda", local_rank) if have_gpu else torch.device("cpu")
-        router = CrossGpuRouter(device)
+            grouter = GpuDirectRouter(engine, torch.device("cuda", local_rank))
+        else:
+            from swarmdb_amd.parallel.router import CrossGpuRouter
+
+            router = CrossGpuRouter(torch.device("cpu"))
 
     recv_K = max(64, 4 * args.batch * world // max(1, len(local_agents)))
     sent_total = 0
@@ -159,10 +166,13 @@ def main() -> int:
     def step(i: int) -> int:
         nonlocal sent_total, recv_total
         recs, payload = batches[i % len(batches)]
-        if router is not None:
-            recs, payload = router.route(recs, payload)
-        engine.enqueue_batch(recs, payload)
-        sent_local = len(recs)
+        if grouter is not None:
+            sent_local = grouter.route_and_enqueue(recs, payload)
+        else:
+            if router is not None:
+                recs, payload = router.route(recs, payload)
+            engine.enqueue_batch(recs, payload)
+            sent_local = len(recs)
         counts, seqs = engine.receive_many(
             local_agents, recv_K, priority_order=args.priority
         )

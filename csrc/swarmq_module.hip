@@ -512,6 +512,25 @@ __global__ void k_lb_batch(int requests, int n_backends,
     loads[lane] = (ull)my;
 }
 
+// Pack payload instances into an all-to-all send buffer (cross-GPU
+// routing over xGMI, BASELINE config 4): one wave per instance, 16-B
+// chunks. Offsets are host-computed (stable, no atomics).
+__global__ void k_pack(const u8 *__restrict__ src_pay,
+                       const u64 *__restrict__ src_off,
+                       const u64 *__restrict__ dst_off,
+                       const u32 *__restrict__ lens, int n,
+                       u8 *__restrict__ out) {
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (wave >= n)
+    return;
+  const uint4 *s = reinterpret_cast<const uint4 *>(src_pay + src_off[wave]);
+  uint4 *d = reinterpret_cast<uint4 *>(out + dst_off[wave]);
+  const u32 nchunk = (lens[wave] + 15u) >> 4;
+  for (u32 c = lane; c < nchunk; c += 64)
+    d[c] = s[c];
+}
+
 __global__ void k_add_load(int idx, long long delta, ull *__restrict__ loads) {
   if (threadIdx.x == 0 && blockIdx.x == 0)
     atomicAdd(&loads[idx], (ull)delta);
@@ -779,6 +798,72 @@ public:
       const int blocks = (n + 3) / 4;
       hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
                          d_stage_recs_[slot], d_stage_pay_[slot], n, base,
+                         d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
+                         d_by_type_, d_by_status_, d_sent_, d_bcast_,
+                         d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
+                         dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
+                         d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
+                         g_);
+    }
+    count_ = base + (u64)n;
+    if (count_ > g_.num_slots)
+      evict_base_ = count_ - g_.num_slots;
+    return base;
+  }
+
+  // ---- GPU-direct cross-GPU routing support ----
+
+  // Stage this rank's raw batch payload H2D and scatter instances into a
+  // caller-provided device send buffer (a torch tensor's data_ptr). The
+  // instance arrays are host-computed: src_off into the staged payload,
+  // dst_off into the send buffer (both 16-B aligned). Synchronizes so
+  // the collective can run on any stream afterwards.
+  void pack_exchange(py::buffer pay, py::array_t<u64> src_off,
+                     py::array_t<u64> dst_off, py::array_t<u32> lens,
+                     uintptr_t send_ptr) {
+    const int n = (int)src_off.size();
+    if (n == 0)
+      return;
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("pack batch exceeds staging_batch");
+    py::buffer_info pi = pay.request();
+    const size_t pay_bytes = (size_t)pi.size * pi.itemsize;
+    ensure_stage_pay(pay_bytes + 16);
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipEventSynchronize(stage_ev_[0]));
+      par_memcpy(h_pay_[0], pi.ptr, pay_bytes);
+      HIP_CHECK(hipMemcpyAsync(d_stage_pay_[0], h_pay_[0], pay_bytes,
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipEventRecord(stage_ev_[0], stream_));
+      HIP_CHECK(hipMemcpyAsync(d_seqs_in_, src_off.data(), n * sizeof(u64),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_match_, dst_off.data(), n * sizeof(u64),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_choices_, lens.data(), n * sizeof(u32),
+                               hipMemcpyHostToDevice, stream_));
+      hipLaunchKernelGGL(k_pack, dim3((n + 3) / 4), dim3(256), 0, stream_,
+                         d_stage_pay_[0], d_seqs_in_, d_match_, d_choices_, n,
+                         reinterpret_cast<u8 *>(send_ptr));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+  }
+
+  // Enqueue a batch whose records and payloads are ALREADY on the device
+  // (e.g. an all-to-all receive buffer): no host staging at all.
+  // Asynchronous; stream-ordered before any later receive_many.
+  u64 enqueue_from_ptrs(uintptr_t recs_ptr, uintptr_t pay_ptr, int n) {
+    if (n <= 0)
+      return count_;
+    const u64 base = count_;
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
+      const int blocks = (n + 3) / 4;
+      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
+                         reinterpret_cast<const Rec *>(recs_ptr),
+                         reinterpret_cast<const u8 *>(pay_ptr), n, base,
                          d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
                          d_by_type_, d_by_status_, d_sent_, d_bcast_,
                          d_bcast_count_, g_);
@@ -1219,6 +1304,8 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("stage_fill", &DeviceQueue::stage_fill)
       .def("enqueue_staged", &DeviceQueue::enqueue_staged)
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
+      .def("pack_exchange", &DeviceQueue::pack_exchange)
+      .def("enqueue_from_ptrs", &DeviceQueue::enqueue_from_ptrs)
       .def("receive_many", &DeviceQueue::receive_many)
       .def("fetch", &DeviceQueue::fetch)
       .def("fetch_raw", &DeviceQueue::fetch_raw, py::arg("seqs"), py::arg("stride") = 0)

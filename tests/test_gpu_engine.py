@@ -444,3 +444,77 @@ def test_group_fanout_parity():
             assert len(rg) == expect
     finally:
         gpu.close()
+
+
+def test_pack_exchange_and_device_ingest(gpu_engine):
+    """GPU-direct routing primitives on one device: pack instances into a
+    torch device buffer (simulating the all-to-all send buffer), then
+    ingest sections straight from device memory. Payloads must survive
+    byte-exact without touching the host."""
+    import torch
+
+    eng = gpu_engine
+    W = 4
+    n_agents = 16
+    for a in range(n_agents):
+        eng.register_agent(a)
+    rng = np.random.default_rng(9)
+    n = 200
+    recs, payload = make_batch(rng, n, n_agents, payload_bytes=96)
+    dest = recs["receiver"].astype(np.int64) % W
+
+    order = np.argsort(dest, kind="stable")
+    inst = order
+    lens = recs["payload_len"][inst].astype(np.uint32)
+    lens16 = ((lens.astype(np.int64) + 15) // 16) * 16
+    dst_off = np.zeros(n, dtype=np.int64)
+    np.cumsum(lens16[:-1], out=dst_off[1:])
+    src_off = recs["payload_off"][inst].astype(np.uint64)
+    counts = np.bincount(dest, minlength=W)
+    bytes_per_rank = np.bincount(dest[inst], weights=lens16.astype(float),
+                                 minlength=W).astype(np.int64)
+    rank_base = np.zeros(W, dtype=np.int64)
+    np.cumsum(bytes_per_rank[:-1], out=rank_base[1:])
+    out_recs = recs[inst].copy()
+    out_recs["payload_off"] = (dst_off - rank_base[dest[inst]]).astype(np.uint64)
+
+    dev = torch.device("cuda", 0)
+    send_pay = torch.empty(int(lens16.sum()), dtype=torch.uint8, device=dev)
+    eng.q.pack_exchange(np.frombuffer(payload, np.uint8), src_off,
+                        dst_off.astype(np.uint64), lens, send_pay.data_ptr())
+    recs_dev = torch.from_numpy(
+        np.frombuffer(out_recs.tobytes(), np.uint8).copy()).to(dev)
+
+    # "exchange" = identity; ingest the W sections from device memory
+    rec_off = pay_off = 0
+    for r in range(W):
+        n_r = int(counts[r])
+        if n_r:
+            eng.q.enqueue_from_ptrs(recs_dev.data_ptr() + rec_off,
+                                    send_pay.data_ptr() + pay_off, n_r)
+        rec_off += n_r * 48
+        pay_off += int(bytes_per_rank[r])
+    eng.q.sync()
+    assert eng.total_messages() == n
+
+    # every message delivered once, payloads byte-exact
+    total = 0
+    src = np.frombuffer(payload, np.uint8)
+    for a in range(n_agents):
+        seqs = eng.receive(a, 1000)
+        total += len(seqs)
+        if len(seqs):
+            hdrs, pays = eng.fetch(seqs[:3])
+            for row, p in zip(hdrs, pays):
+                assert row["receiver"] == a
+                # find original message by matching sender+payload
+            # check payload of first
+    assert total == n
+    # byte-exact check on a sample: fetch everything and match multiset
+    all_seqs = np.arange(n, dtype=np.uint64)
+    hdrs, pays = eng.fetch(all_seqs)
+    got_payloads = sorted(pays)
+    exp_payloads = sorted(
+        src[int(o): int(o) + 96].tobytes() for o in recs["payload_off"]
+    )
+    assert got_payloads == exp_payloads
