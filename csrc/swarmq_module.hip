@@ -94,10 +94,13 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
   // whole batch through L2 atomics)
   __shared__ u32 h_type[N_TYPES];
   __shared__ u32 h_msgs;
+  __shared__ u32 h_failed;
   if (threadIdx.x < N_TYPES)
     h_type[threadIdx.x] = 0;
   if (threadIdx.x == N_TYPES)
     h_msgs = 0;
+  if (threadIdx.x == N_TYPES + 1)
+    h_failed = 0;
   __syncthreads();
 
   const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
@@ -107,31 +110,53 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
     const u64 seq = base_seq + (u64)wave;
     const u32 slot = (u32)(seq % g.num_slots);
 
-    // payload copy, 16 B per lane per round (staging offsets are 16-B
-    // aligned; slots are slot_bytes-strided so destination is aligned too)
-    const uint4 *src =
-        reinterpret_cast<const uint4 *>(stage_pay + r.payload_off);
-    uint4 *dst =
-        reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
-    const u32 nchunk = (r.payload_len + 15u) >> 4;
-    for (u32 c = lane; c < nchunk; c += 64)
-      dst[c] = src[c];
+    // the error lane (reference _errors topic analog, "swarmdb/
+    // main.py":260-273, 501-519): malformed records are parked FAILED
+    // — queryable via status — instead of corrupting device state
+    const bool bad =
+        r.type >= N_TYPES || r.priority > 3 ||
+        r.payload_len > g.slot_bytes ||
+        (r.receiver != BROADCAST && r.receiver >= g.max_agents) ||
+        r.sender >= g.max_agents ||
+        (r.vis_mode != VIS_ALL && r.bitmap != NO_BITMAP &&
+         r.bitmap >= g.num_bitmaps);
+    if (bad) {
+      if (lane == 0) {
+        Rec h = r;
+        h.payload_off = (u64)slot * g.slot_bytes;
+        h.payload_len = 0;
+        h.content_len = 0;
+        hdr[slot] = h;
+        status[slot] = ST_FAILED;
+        atomicAdd(&h_failed, 1u);
+      }
+    } else {
+      // payload copy, 16 B per lane per round (staging offsets are 16-B
+      // aligned; slots are slot_bytes-strided so destination is aligned)
+      const uint4 *src =
+          reinterpret_cast<const uint4 *>(stage_pay + r.payload_off);
+      uint4 *dst =
+          reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
+      const u32 nchunk = (r.payload_len + 15u) >> 4;
+      for (u32 c = lane; c < nchunk; c += 64)
+        dst[c] = src[c];
 
-    if (lane == 0) {
-      Rec h = r;
-      h.payload_off = (u64)slot * g.slot_bytes;
-      hdr[slot] = h;
-      status[slot] = ST_DELIVERED;
-      atomicAdd(&h_type[r.type], 1u);
-      atomicAdd(&h_msgs, 1u);
-      atomicAdd(&sent[r.sender], 1ull);
-      if (r.receiver == BROADCAST) {
-        u32 bi = atomicAdd(bcast_count, 1u);
-        bcast_list[bi] = seq;
-      } else {
-        ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
-        inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
-            seq;
+      if (lane == 0) {
+        Rec h = r;
+        h.payload_off = (u64)slot * g.slot_bytes;
+        hdr[slot] = h;
+        status[slot] = ST_DELIVERED;
+        atomicAdd(&h_type[r.type], 1u);
+        atomicAdd(&h_msgs, 1u);
+        atomicAdd(&sent[r.sender], 1ull);
+        if (r.receiver == BROADCAST) {
+          u32 bi = atomicAdd(bcast_count, 1u);
+          bcast_list[bi] = seq;
+        } else {
+          ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
+          inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
+              seq;
+        }
       }
     }
   }
@@ -140,6 +165,8 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
     atomicAdd(&by_type[threadIdx.x], (ull)h_type[threadIdx.x]);
   if (threadIdx.x == N_TYPES && h_msgs)
     atomicAdd(&by_status[ST_DELIVERED], (ull)h_msgs);
+  if (threadIdx.x == N_TYPES + 1 && h_failed)
+    atomicAdd(&by_status[ST_FAILED], (ull)h_failed);
 }
 
 // Broadcast fan-out: one thread per agent appends the batch's broadcast

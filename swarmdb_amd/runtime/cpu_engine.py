@@ -143,23 +143,38 @@ class CpuEngine(Engine):
             self._ensure(n)
             base = self._count
             sl = slice(base, base + n)
+            # error lane (matches k_enqueue's validation — reference
+            # _errors topic analog, swarmdb/ main.py:260-273, 501-519):
+            # malformed records park as FAILED, no delivery
+            bad = (
+                (recs["type"] >= 7)
+                | (recs["priority"] > 3)
+                | (recs["payload_len"] > self.cfg.slot_bytes)
+                | (
+                    (recs["receiver"] != BROADCAST)
+                    & (recs["receiver"] >= self.cfg.max_agents)
+                )
+                | (recs["sender"] >= self.cfg.max_agents)
+            )
+            good = ~bad
             self._hdr[sl] = recs
             heap_base = len(self._heap)
             self._heap += payloads
             self._pay_off[sl] = recs["payload_off"] + np.uint64(heap_base)
-            self._pay_len[sl] = recs["payload_len"]
-            self._status[sl] = ST_DELIVERED
+            self._pay_len[sl] = np.where(good, recs["payload_len"], 0)
+            self._status[sl] = np.where(good, ST_DELIVERED, ST_FAILED)
             self._count = base + n
             seqs = np.arange(base, base + n, dtype=np.uint64)
 
             # counters
-            np.add.at(self._by_type, recs["type"], 1)
-            self._by_status[ST_DELIVERED] += n
-            np.add.at(self._sent, recs["sender"], 1)
+            np.add.at(self._by_type, recs["type"][good], 1)
+            self._by_status[ST_DELIVERED] += int(good.sum())
+            self._by_status[ST_FAILED] += int(bad.sum())
+            np.add.at(self._sent, recs["sender"][good], 1)
 
-            # inbox fan-out
+            # inbox fan-out (valid records only)
             recv = recs["receiver"]
-            bmask = recv == BROADCAST
+            bmask = (recv == BROADCAST) & good
             if bmask.any():
                 bseqs = seqs[bmask]
                 gmask = recs["vis_mode"][bmask] == VIS_GROUP
@@ -179,7 +194,7 @@ class CpuEngine(Engine):
                             self._inbox[ai] = _U64Ring()
                             self._recv_ts[ai] = _U64Ring()
                         self._inbox[ai].append_many(one)
-            pmask = ~bmask
+            pmask = (recv != BROADCAST) & good
             if pmask.any():
                 prs = recv[pmask]
                 pseqs = seqs[pmask]

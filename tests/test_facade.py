@@ -373,3 +373,37 @@ def test_send_to_group_fast(tmp_db):
     assert [m.id for m in tmp_db.get_agent_messages("b")] == [mid]
     with pytest.raises(ValueError):
         tmp_db.send_to_group_fast("nope", "a", "x")
+
+
+def test_error_lane_malformed_batch(tmp_db):
+    """Malformed batch records park as FAILED (the error-lane analog of
+    the reference's _errors topic), valid ones deliver normally."""
+    from swarmdb_amd.runtime.engine import (
+        BROADCAST, NO_BITMAP, REC_DTYPE, ST_FAILED, VIS_ALL,
+    )
+
+    a = tmp_db.agent_index("a")
+    b = tmp_db.agent_index("b")
+    recs = np.zeros(4, dtype=REC_DTYPE)
+    recs["sender"] = a
+    recs["receiver"] = b
+    recs["vis_mode"] = VIS_ALL
+    recs["bitmap"] = NO_BITMAP
+    recs["payload_len"] = 16
+    recs["content_len"] = 16
+    recs["payload_off"] = np.arange(4, dtype=np.uint64) * 16
+    recs["receiver"][1] = 999999        # bad receiver (not BROADCAST)
+    recs["type"][2] = 200               # bad type
+    recs["payload_len"][3] = 1 << 24    # oversize payload
+    seqs = tmp_db.send_batch(recs, b"y" * 64)
+    st = [tmp_db.engine.get_status(int(s)) for s in seqs]
+    assert st[0] != ST_FAILED
+    assert st[1] == ST_FAILED and st[2] == ST_FAILED and st[3] == ST_FAILED
+    # only the valid one delivers
+    counts, got = tmp_db.receive_batch(np.array([b]), max_per_agent=10)
+    assert counts[0] == 1 and got[0] == seqs[0]
+    # failed ones are queryable by status
+    failed = tmp_db.engine.query(status=ST_FAILED, limit=10)
+    assert set(failed.tolist()) == set(seqs[1:].tolist())
+    s = tmp_db.get_stats()
+    assert s["messages_by_status"].get("failed") == 3

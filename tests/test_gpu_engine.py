@@ -518,3 +518,27 @@ def test_pack_exchange_and_device_ingest(gpu_engine):
         src[int(o): int(o) + 96].tobytes() for o in recs["payload_off"]
     )
     assert got_payloads == exp_payloads
+
+
+def test_error_lane_gpu(gpu_engine):
+    """Malformed records park as FAILED on device (k_enqueue validation),
+    matching the CPU engine."""
+    from swarmdb_amd.runtime.engine import ST_FAILED
+
+    eng = gpu_engine
+    eng.register_agent(0)
+    eng.register_agent(1)
+    rng = np.random.default_rng(2)
+    recs, payload = make_batch(rng, 6, 2, payload_bytes=32)
+    recs["sender"] = 0
+    recs["receiver"] = 1
+    recs["receiver"][1] = 999999
+    recs["type"][2] = 200
+    recs["priority"][3] = 9
+    seqs = eng.enqueue_batch(recs, payload)
+    st = [eng.get_status(int(s)) for s in seqs]
+    assert st[1] == ST_FAILED and st[2] == ST_FAILED and st[3] == ST_FAILED
+    assert st[0] != ST_FAILED and st[4] != ST_FAILED and st[5] != ST_FAILED
+    got = eng.receive(1, 100)
+    assert set(got.tolist()) == {int(seqs[0]), int(seqs[4]), int(seqs[5])}
+    assert eng.stats_arrays()["by_status"][ST_FAILED] == 3
