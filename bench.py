@@ -209,7 +209,8 @@ def main() -> int:
             )
             ndel = int(counts.sum())
             if not args.no_gather and ndel:
-                engine.deliver_payloads(seqs, args.payload)
+                # async: the D2H overlaps the next tick's H2D staging
+                engine.deliver_payloads(seqs, args.payload, synchronize=False)
             sent_total += n_staged
             recv_total += ndel
             _cur[0] = 1 - cur

@@ -994,7 +994,7 @@ public:
   // Gather + D2H into pinned host memory WITHOUT building Python
   // objects — the hot-path delivery step (payload bytes land in host
   // RAM; zero per-message host work). Returns payload bytes landed.
-  u64 fetch_raw(py::array_t<u64> seqs, u32 stride) {
+  u64 fetch_raw(py::array_t<u64> seqs, u32 stride, bool synchronize) {
     const int n = (int)seqs.size();
     if (n == 0)
       return 0;
@@ -1019,11 +1019,20 @@ public:
       HIP_CHECK(hipMemcpyAsync(h_fetch_pay_, d_fetch_pay_,
                                (size_t)n * stride,
                                hipMemcpyDeviceToHost, copy_stream_));
-      HIP_CHECK(hipStreamSynchronize(copy_stream_));
-      for (int i = 0; i < n; ++i)
-        bytes += h_fetch_hdr_[i].payload_len;
+      if (synchronize) {
+        HIP_CHECK(hipStreamSynchronize(copy_stream_));
+        for (int i = 0; i < n; ++i)
+          bytes += h_fetch_hdr_[i].payload_len;
+      } else {
+        bytes = (u64)n * stride; // upper bound; D2H still in flight
+      }
     }
     return bytes;
+  }
+
+  void delivery_sync() {
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipStreamSynchronize(copy_stream_));
   }
 
   void set_status(u64 seq, u32 st) {
@@ -1335,7 +1344,9 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("enqueue_from_ptrs", &DeviceQueue::enqueue_from_ptrs)
       .def("receive_many", &DeviceQueue::receive_many)
       .def("fetch", &DeviceQueue::fetch)
-      .def("fetch_raw", &DeviceQueue::fetch_raw, py::arg("seqs"), py::arg("stride") = 0)
+      .def("fetch_raw", &DeviceQueue::fetch_raw, py::arg("seqs"),
+           py::arg("stride") = 0, py::arg("synchronize") = true)
+      .def("delivery_sync", &DeviceQueue::delivery_sync)
       .def("set_status", &DeviceQueue::set_status)
       .def("get_status", &DeviceQueue::get_status)
       .def("query_range", &DeviceQueue::query_range)

@@ -220,22 +220,34 @@ class GpuEngine(Engine):
             done += chunk
         return out, pays
 
-    def deliver_payloads(self, seqs: np.ndarray, max_payload: int = 0) -> int:
+    def deliver_payloads(
+        self, seqs: np.ndarray, max_payload: int = 0, synchronize: bool = True
+    ) -> int:
         """Gather + D2H the payloads of `seqs` into pinned host memory
         (the delivery step of the hot path) without building per-message
         Python objects. `max_payload` (if known) tightens the D2H stride
-        below slot_bytes. Returns bytes landed."""
+        below slot_bytes. With synchronize=False the D2H stays in flight
+        on the copy stream and overlaps the next tick's H2D (full-duplex
+        PCIe); call delivery_sync() (or any sync'd delivery) to drain.
+        Returns bytes landed (an upper bound when async)."""
         seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
         n = len(seqs)
         done = 0
         total = 0
         while done < n:
             chunk = min(self._staging, n - done)
+            last = done + chunk >= n
             total += int(
-                self.q.fetch_raw(seqs[done : done + chunk], int(max_payload))
+                self.q.fetch_raw(
+                    seqs[done : done + chunk], int(max_payload),
+                    synchronize and last,
+                )
             )
             done += chunk
         return total
+
+    def delivery_sync(self) -> None:
+        self.q.delivery_sync()
 
     def set_status(self, seq: int, status: int) -> None:
         self.q.set_status(int(seq), int(status))
