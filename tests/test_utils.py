@@ -100,3 +100,24 @@ def test_extras_encoding():
     b = wire.encode_extras("id1", {"m": 1}, ["a"])
     d = wire.decode_extras(b)
     assert d == {"id": "id1", "metadata": {"m": 1}, "visible_to": ["a"]}
+
+
+# ---- config ----
+
+def test_queue_config_from_env(monkeypatch):
+    from swarmdb_amd.core.config import QueueConfig
+
+    monkeypatch.setenv("KAFKA_TOPIC_PREFIX", "team_")
+    monkeypatch.setenv("KAFKA_NUM_PARTITIONS", "9")
+    monkeypatch.setenv("MESSAGE_HISTORY_DIR", "/tmp/hist")
+    monkeypatch.setenv("SAVE_INTERVAL_SECONDS", "60")
+    monkeypatch.setenv("SWARMQ_MAX_AGENTS", "256")
+    monkeypatch.setenv("SWARMQ_SLOT_BYTES", "4096")
+    cfg = QueueConfig.from_env(auto_save=False)
+    assert cfg.base_topic == "team_messages"
+    assert cfg.num_partitions == 9
+    assert cfg.save_dir == "/tmp/hist"
+    assert cfg.save_interval == 60.0
+    assert cfg.max_agents == 256
+    assert cfg.slot_bytes == 4096
+    assert cfg.auto_save is False  # override wins
