@@ -57,7 +57,9 @@ class GpuDirectRouter:
         self.group = group
         self.world = dist.get_world_size(group)
         self.rank = dist.get_rank(group)
-        self._hold = None  # keep recv buffers alive across the async enqueue
+        # keep recv buffers alive while async ingest kernels may still
+        # read them (2 ticks deep)
+        self._hold: list = []
 
     def route_and_enqueue(self, recs: np.ndarray, payloads: bytes) -> int:
         W = self.world
@@ -113,17 +115,18 @@ class GpuDirectRouter:
 
         # payload exchange: pack on-device, move over xGMI
         total_out = int(bytes_per_rank.sum())
-        send_pay = torch.empty(max(total_out, 16), dtype=torch.uint8,
+        send_pay = torch.empty(total_out, dtype=torch.uint8,
                                device=self.device)
-        q.pack_exchange(
-            np.frombuffer(payloads, dtype=np.uint8),
-            src_off,
-            dst_off.astype(np.uint64),
-            lens,
-            send_pay.data_ptr(),
-        )
-        recv_pay = torch.empty(max(int(recv_bytes.sum()), 16),
-                               dtype=torch.uint8, device=self.device)
+        if total_out:
+            q.pack_exchange(
+                np.frombuffer(payloads, dtype=np.uint8),
+                src_off,
+                dst_off.astype(np.uint64),
+                lens,
+                send_pay.data_ptr(),
+            )
+        recv_pay = torch.empty(int(recv_bytes.sum()), dtype=torch.uint8,
+                               device=self.device)
         dist.all_to_all_single(
             recv_pay, send_pay,
             output_split_sizes=recv_bytes.tolist(),
@@ -135,7 +138,7 @@ class GpuDirectRouter:
         send_recs = torch.from_numpy(
             np.frombuffer(out_recs.tobytes(), dtype=np.uint8).copy()
         ).to(self.device)
-        recv_recs = torch.empty(max(int(recv_counts.sum()) * REC_BYTES, 16),
+        recv_recs = torch.empty(int(recv_counts.sum()) * REC_BYTES,
                                 dtype=torch.uint8, device=self.device)
         dist.all_to_all_single(
             recv_recs, send_recs,
@@ -160,7 +163,9 @@ class GpuDirectRouter:
                 ingested += n_r
             rec_off += n_r * REC_BYTES
             pay_off += int(recv_bytes[r])
-        self._hold = (recv_recs, recv_pay)  # alive until next tick's sync
+        self._hold.append((recv_recs, recv_pay))
+        if len(self._hold) > 2:
+            self._hold.pop(0)
         return ingested
 
 
