@@ -190,6 +190,61 @@ def config3_group_fanout(seconds: float) -> None:
            "fanout_per_group_msg": 32})
 
 
+def config6_search(seconds: float) -> None:
+    """Bonus: device content search + filtered query over a large
+    HBM-resident log (the reference does Python linear scans,
+    swarmdb/ main.py:671-781)."""
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    n_msgs = 1 << 20  # 1M x 1KB resident messages
+    cfg = QueueConfig(use_gpu=True, auto_save=False, max_agents=1024,
+                      num_slots=n_msgs, slot_bytes=1024 + 64,
+                      staging_batch=1 << 16, inbox_capacity=1 << 12)
+    eng = GpuEngine(cfg)
+    rng = np.random.default_rng(0)
+    agents = np.arange(64, dtype=np.uint32)
+    for a in agents:
+        eng.register_agent(int(a))
+    # fill the log; plant the needle in ~1/4096 messages
+    batch = 1 << 16
+    stride = 1024
+    for b in range(n_msgs // batch):
+        recs, payload = _make_batch(rng, batch, agents, agents, stride)
+        pay = bytearray(payload)
+        for i in range(0, batch, 4096):
+            off = int(recs["payload_off"][i]) + 100
+            pay[off : off + 12] = b"NEEDLE-%05d" % (b % 10)
+        eng.enqueue_batch(recs, bytes(pay))
+    total_bytes = n_msgs * stride
+
+    searches = 0
+    t0 = time.perf_counter()
+    lat = []
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        hits = eng.search(b"NEEDLE-00003", case_sensitive=True, limit=16384)
+        lat.append(time.perf_counter() - s)
+        assert len(hits) > 0
+        searches += 1
+    elapsed = time.perf_counter() - t0
+    qlat = []
+    for _ in range(20):
+        s = time.perf_counter()
+        eng.query(sender=3, type_code=0, limit=10000)
+        qlat.append(time.perf_counter() - s)
+    eng.close()
+    print(json.dumps({
+        "config": 6,
+        "name": "device-content-search",
+        "resident_messages": n_msgs,
+        "resident_payload_gb": round(total_bytes / 1e9, 2),
+        "searches_per_s": round(searches / elapsed, 2),
+        "search_p50_ms": round(float(np.median(lat)) * 1000, 3),
+        "scan_gb_per_s": round(total_bytes * searches / elapsed / 1e9, 1),
+        "query_p50_ms": round(float(np.median(qlat)) * 1000, 3),
+    }))
+
+
 def config5_loadbalancer(seconds: float) -> None:
     """8 mock backends (1 per GPU on a full node); least-loaded dispatch
     via the wavefront min-reduce kernel at >=100k req/s, with concurrent
@@ -243,7 +298,7 @@ def config5_loadbalancer(seconds: float) -> None:
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--config", type=int, required=True,
-                    choices=[1, 2, 3, 4, 5])
+                    choices=[1, 2, 3, 4, 5, 6])
     ap.add_argument("--seconds", type=float, default=5.0)
     args = ap.parse_args()
     if args.config == 1:
@@ -266,6 +321,8 @@ def main() -> int:
         return bench.main()
     elif args.config == 5:
         config5_loadbalancer(args.seconds)
+    elif args.config == 6:
+        config6_search(args.seconds)
     return 0
 
 
