@@ -252,9 +252,13 @@ class CrossGpuRouter:
         dist.all_to_all_single(in_sizes, sizes, group=self.group)
         dist.all_to_all_single(in_nmsgs, nmsgs, group=self.group)
 
-        send_buf = torch.frombuffer(
-            bytearray(b"".join(out_blobs)), dtype=torch.uint8
-        ).to(self.device, non_blocking=False)
+        blob = b"".join(out_blobs)
+        if blob:
+            send_buf = torch.frombuffer(
+                bytearray(blob), dtype=torch.uint8
+            ).to(self.device, non_blocking=False)
+        else:
+            send_buf = torch.empty(0, dtype=torch.uint8, device=self.device)
         total_in = int(in_sizes.sum().item())
         recv_buf = torch.empty(total_in, dtype=torch.uint8, device=self.device)
         dist.all_to_all_single(

@@ -1,0 +1,379 @@
+"""DistributedSwarmsDB — the agent-sharded multi-GPU service.
+
+One instance per rank (one process per GPU, ``torch.distributed``; nccl =
+RCCL over xGMI on GPU, gloo on CPU for CI). Replaces the reference's
+"many workers sharing one Kafka broker" topology (SURVEY.md §2.4 row
+"cross-process shared state") with:
+
+- **control plane, tick-synchronized**: agent registrations, visibility
+  bitmaps and group definitions are queued as ops and applied on EVERY
+  rank in a deterministic order each tick (``all_gather_object``), so the
+  dense agent-index table and bitmap pool stay bit-identical across
+  ranks with no coordinator;
+- **data plane**: outbound messages batch into a per-tick exchange
+  routed by owner rank (``shard_for`` over the stable hash) via
+  all-to-all; an agent's inbox lives only on its owner rank.
+
+Usage::
+
+    svc = DistributedSwarmsDB(config)        # inside an initialized group
+    svc.register_agent("alice")              # queued
+    svc.send_message("alice", "hi", receiver_id="bob")   # queued
+    svc.tick()                               # ALL ranks call together
+    msgs = svc.receive_messages("bob")       # on bob's owner rank
+
+``start_ticker(interval)`` runs the tick loop on a daemon thread when the
+process has no other collectives in flight.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Any, Dict, List, Optional, Union
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..core.config import QueueConfig
+from ..core.message import Message, MessagePriority, MessageStatus, MessageType
+from ..core.wire import encode_content, encode_extras
+from ..runtime.engine import (
+    BROADCAST,
+    FLAG_HAS_EXTRAS,
+    FLAG_JSON_CONTENT,
+    NO_BITMAP,
+    REC_DTYPE,
+    TYPE_CODES,
+    VIS_ALL,
+    VIS_BITMAP,
+)
+from ..runtime.facade import SwarmsDB
+from .router import CrossGpuRouter
+
+
+class DistributedSwarmsDB(SwarmsDB):
+    def __init__(
+        self,
+        config: Optional[QueueConfig] = None,
+        group: Optional[object] = None,
+        **kw,
+    ):
+        self.group = group
+        self.world = dist.get_world_size(group)
+        self.rank = dist.get_rank(group)
+        config = config or QueueConfig()
+        config.world_size = self.world
+        config.rank = self.rank
+        super().__init__(config=config, **kw)
+        self._ctl_ops: List[tuple] = []     # queued control ops
+        self._ctl_seq = 0
+        self._out_msgs: List[Message] = []      # queued sends (local)
+        self._out_recs: List[np.ndarray] = []   # materialized data records
+        self._out_pay: List[bytes] = []
+        self._out_bytes = 0
+        self._pending_meta: Dict[str, Message] = {}  # id -> msg (sender side)
+        device = (
+            torch.device("cuda", config.device_index)
+            if config.use_gpu or (
+                config.use_gpu is None and torch.cuda.is_available()
+            )
+            else torch.device("cpu")
+        )
+        self._router = CrossGpuRouter(device, group)
+        self._ticker: Optional[threading.Thread] = None
+        self._stop = threading.Event()
+
+    # ------------------------------------------------------------------
+    # sharding
+    # ------------------------------------------------------------------
+
+    def owner_rank(self, agent_id: str) -> int:
+        """Owner = dense index mod world — the same mapping the
+        all-to-all router applies to records, and deterministic across
+        ranks because index assignment is tick-synchronized. Requires the
+        agent to be registered (and a tick to have run)."""
+        idx = self._agent_idx.get(agent_id)
+        if idx is None:
+            raise KeyError(
+                f"agent '{agent_id}' not registered yet (register + tick)"
+            )
+        return idx % self.world
+
+    def is_local(self, agent_id: str) -> bool:
+        return self.owner_rank(agent_id) == self.rank
+
+    # ------------------------------------------------------------------
+    # control plane (tick-synchronized)
+    # ------------------------------------------------------------------
+
+    def _queue_ctl(self, op: tuple) -> None:
+        with self._lock:
+            self._ctl_ops.append((self._ctl_seq, *op))
+            self._ctl_seq += 1
+
+    def register_agent(self, agent_id: str) -> bool:  # type: ignore[override]
+        """Queued; the dense index is assigned identically on every rank
+        at the next tick. Idempotent."""
+        with self._lock:
+            if agent_id in self.registered_agents:
+                return True
+        self._queue_ctl(("register", agent_id))
+        return True
+
+    def _apply_register(self, agent_id: str) -> None:
+        with self._lock:
+            if agent_id in self.registered_agents:
+                return
+            idx = self._idx_of(agent_id, create=True)
+            # active (inbox fan-out target) only on the owner rank; the
+            # index table itself is replicated on every rank
+            if self.is_local(agent_id):
+                self.engine.register_agent(idx)
+            self.registered_agents.add(agent_id)
+
+    def deregister_agent(self, agent_id: str) -> bool:  # type: ignore[override]
+        self._queue_ctl(("deregister", agent_id))
+        return True
+
+    def _apply_deregister(self, agent_id: str) -> None:
+        with self._lock:
+            if agent_id not in self.registered_agents:
+                return
+            self.registered_agents.discard(agent_id)
+            idx = self._agent_idx[agent_id]
+            if self.is_local(agent_id):
+                self.engine.deregister_agent(idx)
+
+    def add_agent_group(self, group_name: str, agent_ids: List[str]) -> None:  # type: ignore[override]
+        for a in agent_ids:
+            self.register_agent(a)
+        self._queue_ctl(("group", group_name, list(agent_ids)))
+
+    def _apply_group(self, group_name: str, agent_ids: List[str]) -> None:
+        with self._lock:
+            self.metadata.setdefault("agent_groups", {})[group_name] = list(
+                agent_ids
+            )
+
+    def _apply_bitmap(self, members: List[str]) -> int:
+        """Replicated allocation: every rank allocates the same bitmap at
+        the same pool index (allocators run in lockstep)."""
+        bits = np.zeros(self.config.max_agents, dtype=bool)
+        for a in members:
+            i = self._agent_idx.get(a)
+            if i is not None:
+                bits[i] = True
+        return self.engine.alloc_bitmap(bits)
+
+    # ------------------------------------------------------------------
+    # data plane
+    # ------------------------------------------------------------------
+
+    def send_message(  # type: ignore[override]
+        self,
+        sender_id: str,
+        content: Union[str, Dict[str, Any], List[Any]],
+        receiver_id: Optional[str] = None,
+        message_type: Union[MessageType, str] = MessageType.CHAT,
+        priority: Union[MessagePriority, int] = MessagePriority.NORMAL,
+        metadata: Optional[Dict[str, Any]] = None,
+        visible_to: Optional[List[str]] = None,
+    ) -> str:
+        """Queued send; delivery happens at the next tick on the
+        receiver's owner rank. Returns the message id immediately (the
+        id travels in the payload extras)."""
+        self.register_agent(sender_id)
+        if receiver_id is not None:
+            self.register_agent(receiver_id)
+        mtype = MessageType(message_type)
+        prio = (
+            priority
+            if isinstance(priority, MessagePriority)
+            else MessagePriority(priority)
+        )
+        vis = list(visible_to) if visible_to else []
+        if receiver_id is None and not vis:
+            with self._lock:
+                vis = sorted(self.registered_agents)
+        msg = Message(
+            sender_id=sender_id,
+            receiver_id=receiver_id,
+            content=content,
+            type=mtype,
+            priority=prio,
+            metadata=metadata or {},
+            token_count=self._count_tokens(content),
+            visible_to=vis,
+        )
+        with self._lock:
+            self._out_msgs.append(msg)
+            self._pending_meta[msg.id] = msg
+            if len(self._pending_meta) > 65536:
+                for k in list(self._pending_meta)[:32768]:
+                    del self._pending_meta[k]
+        return msg.id
+
+    def _materialize_send(self, msg: Message,
+                          bitmap_idx: Optional[int] = None) -> None:
+        """Turn a queued Message into a routed record (after control ops
+        of the tick are applied, so indices/bitmaps exist)."""
+        content_b, is_json = encode_content(msg.content)
+        extras_b = encode_extras(msg.id, msg.metadata, msg.visible_to)
+        payload = content_b + extras_b
+        pad = (-len(payload)) % 16
+        payload += b"\x00" * pad
+        rec = np.zeros(1, dtype=REC_DTYPE)
+        rec["sender"] = self._agent_idx[msg.sender_id]
+        if msg.receiver_id is None:
+            rec["receiver"] = BROADCAST
+        else:
+            rec["receiver"] = self._agent_idx[msg.receiver_id]
+        rec["type"] = TYPE_CODES[msg.type.value]
+        rec["priority"] = msg.priority.value
+        rec["timestamp"] = msg.timestamp
+        rec["token_count"] = msg.token_count or 0
+        rec["payload_off"] = self._out_bytes
+        rec["payload_len"] = len(payload) - pad
+        rec["content_len"] = len(content_b)
+        rec["flags"] = FLAG_HAS_EXTRAS | (FLAG_JSON_CONTENT if is_json else 0)
+        if msg.visible_to:
+            rec["vis_mode"] = VIS_BITMAP
+            rec["bitmap"] = bitmap_idx
+        else:
+            rec["vis_mode"] = VIS_ALL
+            rec["bitmap"] = NO_BITMAP
+        self._out_recs.append(rec)
+        self._out_pay.append(payload)
+        self._out_bytes += len(payload)
+
+    # ------------------------------------------------------------------
+    # the tick (ALL ranks must call together)
+    # ------------------------------------------------------------------
+
+    def tick(self) -> int:
+        """One control+data exchange round. Returns the number of
+        messages ingested locally this tick."""
+        with self._lock:
+            my_ops = self._ctl_ops
+            self._ctl_ops = []
+            self._ctl_seq = 0
+            out_msgs, self._out_msgs = self._out_msgs, []
+        # visibility bitmaps must exist in every rank's pool at the same
+        # index: ship the member lists with the control gather and have
+        # EVERY rank allocate them in the same order
+        my_vis = [m.visible_to for m in out_msgs if m.visible_to]
+        gathered: List[Optional[tuple]] = [None] * self.world
+        dist.all_gather_object(gathered, (my_ops, my_vis), group=self.group)
+        gathered_ops = [g[0] if g else [] for g in gathered]
+        gathered_vis = [g[1] if g else [] for g in gathered]
+
+        # apply control ops deterministically: (rank, op_seq) order; all
+        # non-send ops first so sends see a consistent registry
+        all_ops = []
+        for r, ops in enumerate(gathered_ops):
+            for op in ops or []:
+                all_ops.append((r, *op))
+        all_ops.sort(key=lambda t: (t[0], t[1]))
+        for r, _seq, kind, *args in all_ops:
+            if kind == "register":
+                self._apply_register(args[0])
+            elif kind == "deregister":
+                self._apply_deregister(args[0])
+            elif kind == "group":
+                self._apply_group(args[0], args[1])
+        # replicated bitmap allocation (identical pool on every rank)
+        my_bitmaps: List[int] = []
+        for r in range(self.world):
+            for vis_list in gathered_vis[r]:
+                bidx = self._apply_bitmap(vis_list)
+                if r == self.rank:
+                    my_bitmaps.append(bidx)
+        # materialize this rank's queued sends now that the tick's
+        # registry/bitmap state is applied everywhere
+        bit_iter = iter(my_bitmaps)
+        for msg in out_msgs:
+            self._materialize_send(
+                msg, next(bit_iter) if msg.visible_to else None
+            )
+
+        if self._out_recs:
+            recs = np.concatenate(self._out_recs)
+            payload = b"".join(self._out_pay)
+        else:
+            recs = np.empty(0, dtype=REC_DTYPE)
+            payload = b""
+        self._out_recs, self._out_pay, self._out_bytes = [], [], 0
+
+        in_recs, in_pay = self._router.route(recs, payload)
+        if len(in_recs):
+            seqs = self.engine.enqueue_batch(in_recs, in_pay)
+            # map ids for locally-ingested compat-path messages
+            # (ids live in the extras; resolved lazily via fetch)
+            self._maybe_autosave()
+            return len(seqs)
+        return 0
+
+    def flush(self, ticks: int = 2) -> None:
+        """Run `ticks` synchronized rounds (registration + delivery)."""
+        for _ in range(ticks):
+            self.tick()
+
+    def start_ticker(self, interval: float = 0.002) -> None:
+        """Background tick loop (every rank must start one)."""
+        if self._ticker is not None:
+            return
+
+        def loop() -> None:
+            while not self._stop.is_set():
+                self.tick()
+                time.sleep(interval)
+
+        self._ticker = threading.Thread(target=loop, daemon=True)
+        self._ticker.start()
+
+    def stop_ticker(self) -> None:
+        self._stop.set()
+        if self._ticker is not None:
+            self._ticker.join(timeout=10)
+            self._ticker = None
+
+    # ------------------------------------------------------------------
+    # reads are owner-local
+    # ------------------------------------------------------------------
+
+    def receive_messages(  # type: ignore[override]
+        self,
+        agent_id: str,
+        max_messages: int = 100,
+        timeout: float = 1.0,
+        priority_order: bool = False,
+    ):
+        if not self.is_local(agent_id):
+            raise RuntimeError(
+                f"agent '{agent_id}' lives on rank "
+                f"{self.owner_rank(agent_id)}; poll there"
+            )
+        with self._lock:
+            idx = self._agent_idx.get(agent_id)
+        if idx is None:
+            return []
+        deadline = time.monotonic() + max(0.0, timeout)
+        while True:
+            seqs = self.engine.receive(idx, max_messages, priority_order)
+            if len(seqs) or time.monotonic() >= deadline:
+                break
+            time.sleep(0.001)
+        return self._messages_from_seqs(seqs)
+
+    def get_message(self, message_id: str):  # type: ignore[override]
+        m = super().get_message(message_id)
+        if m is not None:
+            return m
+        with self._lock:
+            return self._pending_meta.get(message_id)
+
+    def close(self) -> None:  # type: ignore[override]
+        self.stop_ticker()
+        super().close()
