@@ -961,21 +961,28 @@ class SwarmsDB:
         if older_than_seconds is None:
             older_than_seconds = self.config.retention_ms / 1000.0
         cutoff = time.time() - older_than_seconds
-        seqs = self.engine.query(before=cutoff, limit=1 << 62)
-        if len(seqs) == 0:
+        archive: Dict[str, Any] = {}
+        count = 0
+        # the engine may cap a single query at its staging depth; loop
+        # until the matching set is drained
+        while True:
+            seqs = self.engine.query(before=cutoff, limit=1 << 30)
+            if len(seqs) == 0:
+                break
+            msgs = self._messages_from_seqs(seqs)
+            with self._lock:
+                for s, m in zip(seqs, msgs):
+                    if self.engine.delete(int(s)):
+                        archive[m.id] = m.to_dict()
+                        self._id_to_seq.pop(m.id, None)
+                        count += 1
+        if count == 0:
             return 0
-        msgs = self._messages_from_seqs(seqs)
         archive_dir = self.save_dir / "archives"
         archive_dir.mkdir(parents=True, exist_ok=True)
         path = archive_dir / f"archive_{int(time.time())}.json"
         with open(path, "w") as f:
-            json.dump({m.id: m.to_dict() for m in msgs}, f, indent=2)
-        count = 0
-        with self._lock:
-            for s, m in zip(seqs, msgs):
-                if self.engine.delete(int(s)):
-                    self._id_to_seq.pop(m.id, None)
-                    count += 1
+            json.dump(archive, f, indent=2)
         logger.info("flushed %d old messages to %s", count, path)
         return count
 
