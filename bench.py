@@ -97,18 +97,21 @@ def main() -> int:
     have_gpu = torch.cuda.is_available()
 
     dist_on = world > 1
+    backend = os.environ.get(
+        "SWARMDB_BENCH_BACKEND", "nccl" if have_gpu else "gloo"
+    )
     if dist_on:
         import torch.distributed as dist
 
-        backend = "nccl" if have_gpu else "gloo"
         if have_gpu:
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
         dist.init_process_group(backend=backend)
 
     agents_global = args.agents * world
     max_agents = ((agents_global + 63) // 64) * 64
     rng = np.random.default_rng(1234 + rank)
 
+    n_dev = torch.cuda.device_count() if have_gpu else 1
     cfg = QueueConfig(
         use_gpu=have_gpu,
         max_agents=max_agents,
@@ -116,7 +119,7 @@ def main() -> int:
         slot_bytes=max(2048, ((args.payload + 63) // 16) * 16 + 64),
         inbox_capacity=1 << 16,
         staging_batch=max(16384, args.batch * (2 if dist_on else 1)),
-        device_index=local_rank,
+        device_index=local_rank % max(1, n_dev),
         auto_save=False,
         world_size=world,
         rank=rank,
@@ -140,13 +143,14 @@ def main() -> int:
     router = None
     grouter = None
     if dist_on:
-        if have_gpu:
+        if have_gpu and backend == "nccl":
             # GPU-direct: pack kernel -> RCCL all-to-all over xGMI ->
             # device-side ingest (payloads never touch the host)
             from swarmdb_amd.parallel.router import GpuDirectRouter
 
             grouter = GpuDirectRouter(engine, torch.device("cuda", local_rank))
         else:
+            # host-path exchange (gloo): engine may still be the GPU one
             from swarmdb_amd.parallel.router import CrossGpuRouter
 
             router = CrossGpuRouter(torch.device("cpu"))
@@ -234,11 +238,11 @@ def main() -> int:
         import torch.distributed as dist
 
         t = torch.tensor([elapsed], dtype=torch.float64,
-                         device="cuda" if have_gpu else "cpu")
+                         device="cuda" if (have_gpu and backend == "nccl") else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
         tot = torch.tensor([float(recv_total)], dtype=torch.float64,
-                           device="cuda" if have_gpu else "cpu")
+                           device="cuda" if (have_gpu and backend == "nccl") else "cpu")
         dist.all_reduce(tot, op=dist.ReduceOp.SUM)
         recv_all = int(tot.item())
     else:
