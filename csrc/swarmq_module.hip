@@ -423,56 +423,58 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
                          const u8 *__restrict__ payload,
                          u64 *__restrict__ out, u32 *__restrict__ out_count,
                          u32 cap, QueueGeom g) {
-  // one wave per message: lanes stream the content with coalesced 16-B
-  // loads (1 KiB per wave round) and prefilter on the needle's first
-  // byte; full verification runs only on candidate positions (the
-  // trailing bytes come from L1/L2). ~an order of magnitude over the
-  // byte-at-a-time scan.
-  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  const int lane = threadIdx.x & 63;
-  const int waves = gridDim.x * (blockDim.x >> 6);
+  // Linear-streaming scan: threads stride 16-B chunks of the slot
+  // REGION in seq order — one coalesced uint4 load per chunk, no
+  // per-message header reads on the hot path. Headers/status load only
+  // when a chunk contains the needle's first byte (~6% of random
+  // chunks for 1-in-256 bytes), so the scan runs at HBM streaming rate
+  // instead of header-latency rate. Slot padding bytes are scanned too
+  // and rejected by the content_len bound at verify time.
   const u8 n0 = fold_c(needle[0], fold);
-  for (u64 seq = lo + wave; seq < hi; seq += waves) {
+  const u32 cps = g.slot_bytes >> 4; // chunks per slot
+  const u64 nchunks = (hi - lo) * cps;
+  const u64 stride = (u64)gridDim.x * blockDim.x;
+  for (u64 ci = (u64)blockIdx.x * blockDim.x + threadIdx.x; ci < nchunks;
+       ci += stride) {
+    const u64 seq = lo + ci / cps;
+    const u32 sub = (u32)(ci % cps);
     const u32 slot = (u32)(seq % g.num_slots);
+    const uint4 v = reinterpret_cast<const uint4 *>(
+        payload + (u64)slot * g.slot_bytes)[sub];
+    const u8 *b = reinterpret_cast<const u8 *>(&v);
+    u32 hitmask = 0;
+#pragma unroll
+    for (int j = 0; j < 16; ++j)
+      hitmask |= (fold_c(b[j], fold) == n0) ? (1u << j) : 0u;
+    if (hitmask == 0)
+      continue;
+    // candidate path (rare): check the message and verify positions
+    const Rec h = hdr[slot];
     if (status[slot] == ST_DELETED)
       continue;
-    const Rec h = hdr[slot];
-    if ((int)h.content_len < nlen)
-      continue;
-    const u8 *text = payload + h.payload_off;
     const int clen = (int)h.content_len;
-    const int nstart = clen - nlen + 1;  // valid start positions
+    const int nstart = clen - nlen + 1;
+    if (nstart <= 0)
+      continue;
+    const u8 *text = payload + (u64)slot * g.slot_bytes;
+    const int base = (int)(sub << 4);
     bool found = false;
-    // lane l scans 16-B chunks l, l+64, ... (slot base is 16-B aligned)
-    const int nchunk = (clen + 15) >> 4;
-    for (int c = lane; c < nchunk && !__any(found); c += 64) {
-      const uint4 v = reinterpret_cast<const uint4 *>(text)[c];
-      const u8 *b = reinterpret_cast<const u8 *>(&v);
-      const int base = c << 4;
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int p = base + j;
-        if (p >= nstart)
-          break;
-        if (fold_c(b[j], fold) != n0)
-          continue;
-        bool m = true;
-        // verify from global — the wave just streamed these lines, so
-        // they sit in L1/L2 (avoids dynamic indexing into the register
-        // chunk, which would force scratch)
-        for (int q = 1; q < nlen; ++q) {
-          if (fold_c(text[p + q], fold) != fold_c(needle[q], fold)) {
-            m = false;
-            break;
-          }
-        }
-        if (m) {
-          found = true;
+    while (hitmask && !found) {
+      const int j = __builtin_ctz(hitmask);
+      hitmask &= hitmask - 1;
+      const int p = base + j;
+      if (p >= nstart)
+        continue;
+      bool m = true;
+      for (int q = 1; q < nlen; ++q) {
+        if (fold_c(text[p + q], fold) != fold_c(needle[q], fold)) {
+          m = false;
           break;
         }
       }
+      found = m;
     }
-    if (__any(found) && lane == 0) {
+    if (found) {
       const u32 i = atomicAdd(out_count, 1u);
       if (i < cap)
         out[i] = seq;
@@ -1128,7 +1130,8 @@ public:
                                hipMemcpyHostToDevice, stream_));
       HIP_CHECK(hipMemsetAsync(d_match_count_, 0, sizeof(u32), stream_));
       const u64 span = hi - lo;
-      const int blocks = (int)std::min<u64>((span + 3) / 4, 2048);
+      const u64 chunks = span * (g_.slot_bytes >> 4);
+      const int blocks = (int)std::min<u64>((chunks + 255) / 256, 2048);
       hipLaunchKernelGGL(k_search, dim3(blocks), dim3(256), 0, stream_, lo, hi,
                          d_needle_, (int)nd.size(), fold ? 1 : 0, d_hdr_,
                          d_status_, d_payload_, d_match_, d_match_count_, cap,

@@ -321,7 +321,9 @@ class GpuEngine(Engine):
             seqs = self.q.search_range(
                 lo, hi, needle, not case_sensitive, self._staging
             )
-            seqs = np.sort(np.asarray(seqs, dtype=np.uint64))[::-1]
+            # the linear-scan kernel may report a message once per
+            # matching chunk: dedup
+            seqs = np.unique(np.asarray(seqs, dtype=np.uint64))[::-1]
             found.append(seqs)
             nfound += len(seqs)
             hi = lo
