@@ -515,6 +515,17 @@ __global__ void k_unread(const u32 *__restrict__ agents, int n_agents,
     out[b] = cnt;
 }
 
+// Batched status gather (no payload traffic).
+__global__ void k_statuses(const u64 *__restrict__ seqs, int n,
+                           u64 evict_base, const u32 *__restrict__ status,
+                           u32 *__restrict__ out, QueueGeom g) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n)
+    return;
+  const u64 s = seqs[i];
+  out[i] = (s < evict_base) ? ST_DELETED : status[s % g.num_slots];
+}
+
 // Status mutation with counter upkeep (mark processed / admin status
 // updates / delete tombstones).
 __global__ void k_set_status(u64 seq, u32 new_status, u32 *__restrict__ status,
@@ -1061,6 +1072,31 @@ public:
     HIP_CHECK(hipStreamSynchronize(copy_stream_));
   }
 
+  py::array_t<u32> get_statuses(py::array_t<u64> seqs) {
+    const int n = (int)seqs.size();
+    py::array_t<u32> out(n);
+    if (n == 0)
+      return out;
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("status batch exceeds staging_batch");
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipEventRecord(ev_, stream_));
+      HIP_CHECK(hipStreamWaitEvent(copy_stream_, ev_, 0));
+      HIP_CHECK(hipMemcpyAsync(d_seqs_in_, seqs.data(), n * sizeof(u64),
+                               hipMemcpyHostToDevice, copy_stream_));
+      hipLaunchKernelGGL(k_statuses, dim3((n + 255) / 256), dim3(256), 0,
+                         copy_stream_, d_seqs_in_, n, evict_base_, d_status_,
+                         d_fetch_status_, g_);
+      HIP_CHECK(hipMemcpyAsync(h_fetch_status_, d_fetch_status_,
+                               n * sizeof(u32), hipMemcpyDeviceToHost,
+                               copy_stream_));
+      HIP_CHECK(hipStreamSynchronize(copy_stream_));
+    }
+    std::memcpy(out.mutable_data(), h_fetch_status_, n * sizeof(u32));
+    return out;
+  }
+
   void set_status(u64 seq, u32 st) {
     hipLaunchKernelGGL(k_set_status, dim3(1), dim3(64), 0, stream_, seq, st,
                        d_status_, d_by_status_, g_);
@@ -1376,6 +1412,7 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("delivery_sync", &DeviceQueue::delivery_sync)
       .def("set_status", &DeviceQueue::set_status)
       .def("get_status", &DeviceQueue::get_status)
+      .def("get_statuses", &DeviceQueue::get_statuses)
       .def("query_range", &DeviceQueue::query_range)
       .def("search_range", &DeviceQueue::search_range)
       .def("inbox_window", &DeviceQueue::inbox_window)

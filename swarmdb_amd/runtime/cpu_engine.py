@@ -340,6 +340,10 @@ class CpuEngine(Engine):
     def get_status(self, seq: int) -> int:
         return int(self._status[seq])
 
+    def statuses(self, seqs: np.ndarray) -> np.ndarray:
+        with self._lock:
+            return self._status[np.asarray(seqs, dtype=np.int64)].copy()
+
     def query(
         self,
         sender: Optional[int] = None,

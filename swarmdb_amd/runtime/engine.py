@@ -184,6 +184,13 @@ class Engine(abc.ABC):
     @abc.abstractmethod
     def get_status(self, seq: int) -> int: ...
 
+    def statuses(self, seqs: np.ndarray) -> np.ndarray:
+        """Batched status lookup (one kernel + one D2H on GPU)."""
+        return np.fromiter(
+            (self.get_status(int(s)) for s in seqs), dtype=np.uint8,
+            count=len(seqs),
+        )
+
     @abc.abstractmethod
     def query(
         self,

@@ -613,13 +613,9 @@ class SwarmsDB:
             if idx is None:
                 return []
         seqs = self.engine.peek_inbox(idx)[::-1]  # newest first
-        if status is not None:
+        if status is not None and len(seqs):
             code = _status_code(status)
-            st = np.fromiter(
-                (self.engine.get_status(int(s)) for s in seqs),
-                dtype=np.int64,
-                count=len(seqs),
-            )
+            st = self.engine.statuses(seqs)
             seqs = seqs[st == code]
         seqs = seqs[skip : skip + limit]
         return self._messages_from_seqs(seqs.copy())
@@ -827,12 +823,10 @@ class SwarmsDB:
         """The §2.1 history schema (reference swarmdb/ main.py:877-884)."""
         n = self.engine.total_messages()
         seqs = np.arange(n, dtype=np.uint64)
-        msgs = self._messages_from_seqs(seqs) if n else []
-        alive = {
-            int(s): m
-            for s, m in zip(seqs, msgs)
-            if self.engine.get_status(int(s)) != ST_DELETED
-        }
+        st = self.engine.statuses(seqs) if n else np.empty(0, dtype=np.uint8)
+        seqs = seqs[st != ST_DELETED]
+        msgs = self._messages_from_seqs(seqs) if len(seqs) else []
+        alive = {int(s): m for s, m in zip(seqs, msgs)}
         seq_to_id = {s: m.id for s, m in alive.items()}
         with self._lock:
             inbox_obj: Dict[str, List[str]] = {}

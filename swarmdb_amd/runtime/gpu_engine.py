@@ -188,8 +188,7 @@ class GpuEngine(Engine):
         return entries[st != ST_DELETED]
 
     def _statuses(self, seqs: np.ndarray) -> np.ndarray:
-        hdrs, _ = self.fetch(seqs)
-        return hdrs["status"]
+        return self.statuses(seqs)
 
     def unread_count(self, agent_idx: int) -> int:
         out = self.q.unread_counts(np.array([agent_idx], dtype=np.uint32))
@@ -254,6 +253,18 @@ class GpuEngine(Engine):
 
     def get_status(self, seq: int) -> int:
         return int(self.q.get_status(int(seq)))
+
+    def statuses(self, seqs: np.ndarray) -> np.ndarray:
+        seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
+        out = np.empty(len(seqs), dtype=np.uint8)
+        done = 0
+        while done < len(seqs):
+            chunk = min(self._staging, len(seqs) - done)
+            out[done : done + chunk] = self.q.get_statuses(
+                seqs[done : done + chunk]
+            ).astype(np.uint8)
+            done += chunk
+        return out
 
     def query(
         self,
