@@ -208,6 +208,9 @@ def main() -> int:
             nr, npay = batches[(i + 1) % len(batches)]
             q.stage_fill(1 - cur, nr, np.frombuffer(npay, dtype=np.uint8),
                          len(nr))
+            # upload batch i+1 on the H2D stream: overlaps batch i's
+            # kernels and the delivery D2H (full-duplex PCIe)
+            q.prefetch_staged(1 - cur)
             counts, seqs = engine.receive_many(
                 local_agents, recv_K, priority_order=args.priority
             )

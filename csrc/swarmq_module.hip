@@ -633,7 +633,10 @@ public:
     HIP_CHECK(hipSetDevice(device_));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&h2d_stream_, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&ev_, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&up_ev_[0], hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&up_ev_[1], hipEventDisableTiming));
 
     // device state
     HIP_CHECK(hipMalloc(&d_hdr_, (size_t)num_slots * sizeof(Rec)));
@@ -735,8 +738,11 @@ public:
     (void)hipEventDestroy(stage_ev_[0]);
     (void)hipEventDestroy(stage_ev_[1]);
     (void)hipEventDestroy(ev_);
+    (void)hipEventDestroy(up_ev_[0]);
+    (void)hipEventDestroy(up_ev_[1]);
     (void)hipStreamDestroy(stream_);
     (void)hipStreamDestroy(copy_stream_);
+    (void)hipStreamDestroy(h2d_stream_);
   }
 
   // ---- registry ----
@@ -845,6 +851,28 @@ public:
     staged_pay_[slot] = pay_bytes;
   }
 
+  // Upload a filled slot's staging buffers on the dedicated H2D stream
+  // (overlaps the main stream's kernels and the copy stream's delivery
+  // D2H — PCIe is full duplex). enqueue_staged picks the upload up.
+  void prefetch_staged(int slot) {
+    if (slot < 0 || slot > 1)
+      throw std::invalid_argument("slot must be 0 or 1");
+    const int n = staged_n_[slot];
+    if (n <= 0)
+      return;
+    const size_t pay_bytes = staged_pay_[slot];
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot], h_recs_[slot],
+                             (size_t)n * sizeof(Rec), hipMemcpyHostToDevice,
+                             h2d_stream_));
+    if (pay_bytes)
+      HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot], h_pay_[slot], pay_bytes,
+                               hipMemcpyHostToDevice, h2d_stream_));
+    HIP_CHECK(hipEventRecord(stage_ev_[slot], h2d_stream_));
+    HIP_CHECK(hipEventRecord(up_ev_[slot], h2d_stream_));
+    uploaded_[slot] = true;
+  }
+
   // Launch H2D + enqueue kernels for a previously filled slot; returns
   // the base seq. NO sync — receive_many on the same stream is ordered
   // after it.
@@ -858,13 +886,19 @@ public:
     const u64 base = count_;
     {
       py::gil_scoped_release nogil;
-      HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot], h_recs_[slot],
-                               (size_t)n * sizeof(Rec),
-                               hipMemcpyHostToDevice, stream_));
-      if (pay_bytes)
-        HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot], h_pay_[slot], pay_bytes,
+      if (uploaded_[slot]) {
+        // already uploaded by prefetch_staged: order kernels after it
+        HIP_CHECK(hipStreamWaitEvent(stream_, up_ev_[slot], 0));
+        uploaded_[slot] = false;
+      } else {
+        HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot], h_recs_[slot],
+                                 (size_t)n * sizeof(Rec),
                                  hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipEventRecord(stage_ev_[slot], stream_));
+        if (pay_bytes)
+          HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot], h_pay_[slot],
+                                   pay_bytes, hipMemcpyHostToDevice, stream_));
+        HIP_CHECK(hipEventRecord(stage_ev_[slot], stream_));
+      }
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
       const int blocks = (n + 3) / 4;
       hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
@@ -1332,8 +1366,10 @@ private:
   size_t out_pool_ = 0;
   size_t stage_pay_bytes_ = 0;
 
-  hipStream_t stream_{}, copy_stream_{};
+  hipStream_t stream_{}, copy_stream_{}, h2d_stream_{};
   hipEvent_t ev_{};
+  hipEvent_t up_ev_[2] = {};
+  bool uploaded_[2] = {false, false};
 
   Rec *d_hdr_{};
   u32 *d_status_{};
@@ -1409,6 +1445,7 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("sync", &DeviceQueue::sync)
       .def("stage_fill", &DeviceQueue::stage_fill)
       .def("enqueue_staged", &DeviceQueue::enqueue_staged)
+      .def("prefetch_staged", &DeviceQueue::prefetch_staged)
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
       .def("pack_exchange", &DeviceQueue::pack_exchange)
       .def("enqueue_from_ptrs", &DeviceQueue::enqueue_from_ptrs)

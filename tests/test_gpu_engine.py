@@ -393,6 +393,8 @@ def test_staged_pipeline_roundtrip(gpu_engine):
         recs["receiver"] = 1
         slot = it % 2
         eng.q.stage_fill(slot, recs, np.frombuffer(payload, np.uint8), n)
+        if it % 2:
+            eng.q.prefetch_staged(slot)  # exercise the H2D-prefetch path
         base = eng.q.enqueue_staged(slot)
         src = np.frombuffer(payload, np.uint8)
         for j in (0, n - 1):
