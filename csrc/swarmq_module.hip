@@ -533,6 +533,25 @@ __global__ void k_statuses(const u64 *__restrict__ seqs, int n,
   out[i] = (s < evict_base) ? ST_DELETED : status[s % g.num_slots];
 }
 
+// Batched status mutation with counter upkeep (history load restore).
+__global__ void k_set_statuses(const u64 *__restrict__ seqs,
+                               const u32 *__restrict__ new_status, int n,
+                               u32 *__restrict__ status,
+                               ull *__restrict__ by_status, QueueGeom g) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n)
+    return;
+  const u32 slot = (u32)(seqs[i] % g.num_slots);
+  const u32 ns = new_status[i];
+  const u32 old = atomicExch(&status[slot], ns);
+  if (old != ns) {
+    if (old < N_STATUS)
+      atomicAdd(&by_status[old], (ull)(-1ll));
+    if (ns < N_STATUS)
+      atomicAdd(&by_status[ns], 1ull);
+  }
+}
+
 // Status mutation with counter upkeep (mark processed / admin status
 // updates / delete tombstones).
 __global__ void k_set_status(u64 seq, u32 new_status, u32 *__restrict__ status,
@@ -1138,6 +1157,23 @@ public:
     return out;
   }
 
+  void set_statuses(py::array_t<u64> seqs, py::array_t<u32> sts) {
+    const int n = (int)seqs.size();
+    if (n == 0)
+      return;
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("status batch exceeds staging_batch");
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipMemcpyAsync(d_seqs_in_, seqs.data(), n * sizeof(u64),
+                             hipMemcpyHostToDevice, stream_));
+    HIP_CHECK(hipMemcpyAsync(d_choices_, sts.data(), n * sizeof(u32),
+                             hipMemcpyHostToDevice, stream_));
+    hipLaunchKernelGGL(k_set_statuses, dim3((n + 255) / 256), dim3(256), 0,
+                       stream_, d_seqs_in_, d_choices_, n, d_status_,
+                       d_by_status_, g_);
+    HIP_CHECK(hipStreamSynchronize(stream_));
+  }
+
   void set_status(u64 seq, u32 st) {
     hipLaunchKernelGGL(k_set_status, dim3(1), dim3(64), 0, stream_, seq, st,
                        d_status_, d_by_status_, g_);
@@ -1455,6 +1491,7 @@ PYBIND11_MODULE(_swarmq, m) {
            py::arg("stride") = 0, py::arg("synchronize") = true)
       .def("delivery_sync", &DeviceQueue::delivery_sync)
       .def("set_status", &DeviceQueue::set_status)
+      .def("set_statuses", &DeviceQueue::set_statuses)
       .def("get_status", &DeviceQueue::get_status)
       .def("get_statuses", &DeviceQueue::get_statuses)
       .def("query_range", &DeviceQueue::query_range)

@@ -344,6 +344,14 @@ class CpuEngine(Engine):
         with self._lock:
             return self._status[np.asarray(seqs, dtype=np.int64)].copy()
 
+    def set_statuses(self, seqs: np.ndarray, statuses: np.ndarray) -> None:
+        with self._lock:
+            idx = np.asarray(seqs, dtype=np.int64)
+            old = self._status[idx]
+            np.add.at(self._by_status, old, -1)
+            np.add.at(self._by_status, statuses, 1)
+            self._status[idx] = statuses
+
     def query(
         self,
         sender: Optional[int] = None,

@@ -181,6 +181,11 @@ class Engine(abc.ABC):
     @abc.abstractmethod
     def set_status(self, seq: int, status: int) -> None: ...
 
+    def set_statuses(self, seqs: np.ndarray, statuses: np.ndarray) -> None:
+        """Batched status restore (one kernel on GPU)."""
+        for s, st in zip(seqs, statuses):
+            self.set_status(int(s), int(st))
+
     @abc.abstractmethod
     def get_status(self, seq: int) -> int: ...
 

@@ -907,36 +907,54 @@ class SwarmsDB:
                 if m.receiver_id is not None:
                     self.register_agent(m.receiver_id)
                 self.register_agent(m.sender_id)
-            for m in msgs:
-                rec = np.zeros(1, dtype=REC_DTYPE)
+            # one batched enqueue + one batched status restore (a
+            # per-message replay would cost a kernel launch per message)
+            n = len(msgs)
+            recs = np.zeros(n, dtype=REC_DTYPE)
+            chunks: List[bytes] = []
+            off = 0
+            for i, m in enumerate(msgs):
                 content_b, is_json = encode_content(m.content)
                 extras_b = encode_extras(m.id, m.metadata, m.visible_to)
                 payload = content_b + extras_b
-                rec["sender"] = self._agent_idx[m.sender_id]
-                rec["receiver"] = (
+                pad = (-len(payload)) % 16
+                chunks.append(payload + b"\x00" * pad)
+                recs["sender"][i] = self._agent_idx[m.sender_id]
+                recs["receiver"][i] = (
                     BROADCAST
                     if m.receiver_id is None
                     else self._agent_idx[m.receiver_id]
                 )
-                rec["type"] = _type_code(m.type)
-                rec["priority"] = m.priority.value
-                rec["timestamp"] = m.timestamp
-                rec["token_count"] = m.token_count or 0
-                rec["payload_len"] = len(payload)
-                rec["content_len"] = len(content_b)
-                rec["flags"] = FLAG_HAS_EXTRAS | (
+                recs["type"][i] = _type_code(m.type)
+                recs["priority"][i] = m.priority.value
+                recs["timestamp"][i] = m.timestamp
+                recs["token_count"][i] = m.token_count or 0
+                recs["payload_off"][i] = off
+                recs["payload_len"][i] = len(payload)
+                recs["content_len"][i] = len(content_b)
+                recs["flags"][i] = FLAG_HAS_EXTRAS | (
                     FLAG_JSON_CONTENT if is_json else 0
                 )
                 if m.visible_to:
-                    rec["vis_mode"] = VIS_BITMAP
-                    rec["bitmap"] = self._bitmap_for(m.visible_to)
+                    recs["vis_mode"][i] = VIS_BITMAP
+                    recs["bitmap"][i] = self._bitmap_for(m.visible_to)
                 else:
-                    rec["vis_mode"] = VIS_ALL
-                    rec["bitmap"] = NO_BITMAP
-                seq = int(self.engine.enqueue_batch(rec, payload)[0])
-                self._id_to_seq[m.id] = seq
-                self.engine.set_status(seq, _status_code(m.status))
-            return len(msgs)
+                    recs["vis_mode"][i] = VIS_ALL
+                    recs["bitmap"][i] = NO_BITMAP
+                off += len(payload) + pad
+            if n:
+                seqs = self.engine.enqueue_batch(recs, b"".join(chunks))
+                for m, s in zip(msgs, seqs):
+                    self._id_to_seq[m.id] = int(s)
+                self.engine.set_statuses(
+                    seqs,
+                    np.fromiter(
+                        (_status_code(m.status) for m in msgs),
+                        dtype=np.uint32,
+                        count=n,
+                    ),
+                )
+            return n
 
     def export_as_yaml(self) -> str:
         """History object as YAML (reference swarmdb/ main.py:936-971)."""

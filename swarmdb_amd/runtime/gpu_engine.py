@@ -254,6 +254,17 @@ class GpuEngine(Engine):
     def get_status(self, seq: int) -> int:
         return int(self.q.get_status(int(seq)))
 
+    def set_statuses(self, seqs: np.ndarray, statuses: np.ndarray) -> None:
+        seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
+        statuses = np.ascontiguousarray(statuses, dtype=np.uint32)
+        done = 0
+        while done < len(seqs):
+            chunk = min(self._staging, len(seqs) - done)
+            self.q.set_statuses(
+                seqs[done : done + chunk], statuses[done : done + chunk]
+            )
+            done += chunk
+
     def statuses(self, seqs: np.ndarray) -> np.ndarray:
         seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
         out = np.empty(len(seqs), dtype=np.uint8)
