@@ -431,7 +431,6 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
   // instead of header-latency rate. Slot padding bytes are scanned too
   // and rejected by the content_len bound at verify time.
   const u8 n0 = fold_c(needle[0], fold);
-  const u8 n1 = nlen > 1 ? fold_c(needle[1], fold) : 0;
   const u32 cps = g.slot_bytes >> 4; // chunks per slot
   const u64 nchunks = (hi - lo) * cps;
   const u64 stride = (u64)gridDim.x * blockDim.x;
@@ -443,16 +442,10 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
     const uint4 v = reinterpret_cast<const uint4 *>(
         payload + (u64)slot * g.slot_bytes)[sub];
     const u8 *b = reinterpret_cast<const u8 *>(&v);
-    // two-byte prefilter in registers (position 15's second byte lives
-    // in the next chunk — keep it as a candidate)
     u32 hitmask = 0;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const bool h0 = fold_c(b[j], fold) == n0;
-      const bool h1 =
-          nlen <= 1 || j == 15 || fold_c(b[j + 1], fold) == n1;
-      hitmask |= (h0 && h1) ? (1u << j) : 0u;
-    }
+    for (int j = 0; j < 16; ++j)
+      hitmask |= (fold_c(b[j], fold) == n0) ? (1u << j) : 0u;
     if (hitmask == 0)
       continue;
     // candidate path (rare): check the message and verify positions
