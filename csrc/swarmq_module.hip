@@ -863,6 +863,46 @@ public:
     staged_pay_[slot] = pay_bytes;
   }
 
+  // Allocate pinned host memory exposed as a numpy array (freed by the
+  // array's capsule). Producers can build batches in-place and hand the
+  // pointers to prefetch_from — zero host-side copies on the send path.
+  static py::array alloc_pinned(size_t nbytes) {
+    void *p = nullptr;
+    HIP_CHECK(hipHostMalloc(&p, nbytes));
+    py::capsule owner(p, [](void *q) { (void)hipHostFree(q); });
+    return py::array_t<u8>({(py::ssize_t)nbytes}, {(py::ssize_t)1},
+                           static_cast<u8 *>(p), owner);
+  }
+
+  // Upload a batch that ALREADY lives in pinned host memory (e.g. from
+  // alloc_pinned) straight onto the H2D stream — no staging memcpy.
+  // The caller must not rewrite the buffers until the upload completes
+  // (one tick later with double-buffered slots).
+  void prefetch_from(int slot, uintptr_t recs_ptr, uintptr_t pay_ptr, int n,
+                     size_t pay_bytes) {
+    if (slot < 0 || slot > 1)
+      throw std::invalid_argument("slot must be 0 or 1");
+    if ((u32)n > staging_batch_)
+      throw std::invalid_argument("batch exceeds staging_batch");
+    ensure_stage_pay(pay_bytes + 16);
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot],
+                               reinterpret_cast<void *>(recs_ptr),
+                               (size_t)n * sizeof(Rec),
+                               hipMemcpyHostToDevice, h2d_stream_));
+      if (pay_bytes)
+        HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot],
+                                 reinterpret_cast<void *>(pay_ptr), pay_bytes,
+                                 hipMemcpyHostToDevice, h2d_stream_));
+      HIP_CHECK(hipEventRecord(stage_ev_[slot], h2d_stream_));
+      HIP_CHECK(hipEventRecord(up_ev_[slot], h2d_stream_));
+    }
+    staged_n_[slot] = n;
+    staged_pay_[slot] = pay_bytes;
+    uploaded_[slot] = true;
+  }
+
   // Upload a filled slot's staging buffers on the dedicated H2D stream
   // (overlaps the main stream's kernels and the copy stream's delivery
   // D2H — PCIe is full duplex). enqueue_staged picks the upload up.
@@ -1475,6 +1515,8 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("stage_fill", &DeviceQueue::stage_fill)
       .def("enqueue_staged", &DeviceQueue::enqueue_staged)
       .def("prefetch_staged", &DeviceQueue::prefetch_staged)
+      .def("prefetch_from", &DeviceQueue::prefetch_from)
+      .def_static("alloc_pinned", &DeviceQueue::alloc_pinned)
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
       .def("pack_exchange", &DeviceQueue::pack_exchange)
       .def("enqueue_from_ptrs", &DeviceQueue::enqueue_from_ptrs)
