@@ -819,15 +819,62 @@ class SwarmsDB:
     # 1327-1365)
     # ------------------------------------------------------------------
 
+    def _message_dict(self, row: np.void, payload: bytes) -> Dict[str, Any]:
+        """Reconstruct the wire dict straight from an engine row —
+        the spill path skips pydantic construction entirely."""
+        clen = int(row["content_len"])
+        flags = int(row["flags"])
+        extras = (
+            decode_extras(payload[clen:]) if flags & FLAG_HAS_EXTRAS else {}
+        )
+        seq = int(row["seq"])
+        recv = int(row["receiver"])
+        sidx = int(row["sender"])
+        return {
+            "id": extras.get("id") or derived_id(self.config.rank, seq),
+            "sender_id": (
+                self._agent_ids[sidx]
+                if sidx < len(self._agent_ids)
+                else f"agent{sidx}"
+            ),
+            "receiver_id": (
+                None
+                if recv == BROADCAST
+                else (
+                    self._agent_ids[recv]
+                    if recv < len(self._agent_ids)
+                    else f"agent{recv}"
+                )
+            ),
+            "content": decode_content(
+                payload[:clen], bool(flags & FLAG_JSON_CONTENT)
+            ),
+            "type": TYPE_NAMES[int(row["type"])],
+            "priority": int(row["priority"]),
+            "timestamp": float(row["timestamp"]),
+            "status": STATUS_NAMES[min(int(row["status"]), ST_FAILED)],
+            "metadata": extras.get("metadata", {}),
+            "token_count": int(row["token_count"]),
+            "visible_to": extras.get("visible_to", []),
+        }
+
     def _history_object(self) -> Dict[str, Any]:
         """The §2.1 history schema (reference swarmdb/ main.py:877-884)."""
         n = self.engine.total_messages()
         seqs = np.arange(n, dtype=np.uint64)
         st = self.engine.statuses(seqs) if n else np.empty(0, dtype=np.uint8)
         seqs = seqs[st != ST_DELETED]
-        msgs = self._messages_from_seqs(seqs) if len(seqs) else []
-        alive = {int(s): m for s, m in zip(seqs, msgs)}
-        seq_to_id = {s: m.id for s, m in alive.items()}
+        messages: Dict[str, Any] = {}
+        seq_to_id: Dict[int, str] = {}
+        done = 0
+        while done < len(seqs):
+            chunk = seqs[done : done + 16384]
+            hdrs, payloads = self.engine.fetch(chunk)
+            for row, payload in zip(hdrs, payloads):
+                d = self._message_dict(row, payload)
+                messages[d["id"]] = d
+                seq_to_id[int(row["seq"])] = d["id"]
+            done += len(chunk)
         with self._lock:
             inbox_obj: Dict[str, List[str]] = {}
             for agent_id, idx in self._agent_idx.items():
@@ -837,7 +884,7 @@ class SwarmsDB:
                 ]
             registered = sorted(self.registered_agents)
         return {
-            "messages": {m.id: m.to_dict() for m in alive.values()},
+            "messages": messages,
             "agent_inbox": inbox_obj,
             "registered_agents": registered,
             "timestamp": time.time(),
