@@ -30,7 +30,7 @@ from fastapi.responses import JSONResponse
 from fastapi.security import HTTPAuthorizationCredentials, HTTPBearer
 
 from ..core.config import QueueConfig
-from ..core.message import MessagePriority, MessageStatus, MessageType
+from ..core.message import MessageStatus, MessageType
 from ..runtime.facade import SwarmsDB
 from ..utils import jwt as jwtlib
 from .models import (

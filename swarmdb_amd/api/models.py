@@ -4,9 +4,9 @@ from __future__ import annotations
 
 from typing import Any, Dict, List, Optional, Union
 
-from pydantic import BaseModel, Field
+from pydantic import BaseModel
 
-from ..core.message import Message, MessagePriority, MessageStatus, MessageType
+from ..core.message import Message, MessagePriority, MessageType
 
 
 class UserCredentials(BaseModel):

@@ -10,7 +10,7 @@ tier (ring sizes, slot bytes, staging depth) is new — sized for MI355X
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

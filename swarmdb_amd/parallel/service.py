@@ -37,7 +37,7 @@ import torch
 import torch.distributed as dist
 
 from ..core.config import QueueConfig
-from ..core.message import Message, MessagePriority, MessageStatus, MessageType
+from ..core.message import Message, MessagePriority, MessageType
 from ..core.wire import encode_content, encode_extras
 from ..runtime.engine import (
     BROADCAST,

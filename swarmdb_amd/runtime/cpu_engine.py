@@ -29,9 +29,7 @@ from .engine import (
     ST_DELIVERED,
     ST_FAILED,
     ST_PENDING,
-    ST_PROCESSED,
     ST_READ,
-    VIS_ALL,
     VIS_BITMAP,
     VIS_GROUP,
     Engine,

@@ -23,10 +23,8 @@ from __future__ import annotations
 
 import json
 import logging
-import os
 import threading
 import time
-import uuid
 from concurrent.futures import ThreadPoolExecutor
 from datetime import datetime
 from pathlib import Path
@@ -49,17 +47,13 @@ from ..utils.hashing import partition_for
 from ..utils.tracing import tracer
 from .engine import (
     BROADCAST,
-    FLAG_DERIVED_ID,
     FLAG_HAS_EXTRAS,
     FLAG_JSON_CONTENT,
     NO_BITMAP,
     REC_DTYPE,
     ST_DELETED,
-    ST_DELIVERED,
     ST_FAILED,
-    ST_PENDING,
     ST_PROCESSED,
-    ST_READ,
     STATUS_NAMES,
     TYPE_CODES,
     TYPE_NAMES,

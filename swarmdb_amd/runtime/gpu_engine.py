@@ -21,7 +21,6 @@ from ..core.config import QueueConfig
 from .engine import (
     REC_DTYPE,
     ST_DELETED,
-    ST_DELIVERED,
     Engine,
 )
 

@@ -12,7 +12,7 @@ import hashlib
 import hmac
 import json
 import time
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 
 class JWTError(Exception):
