@@ -87,6 +87,8 @@ def main() -> int:
                     help="priority-ordered dequeue (config 3 kernel)")
     ap.add_argument("--no-gather", action="store_true",
                     help="skip payload D2H gather (delivery stays device-side)")
+    ap.add_argument("--dump-steps", action="store_true",
+                    help="print per-step wall times (variance diagnosis)")
     args = ap.parse_args()
 
     import torch
@@ -268,6 +270,9 @@ def main() -> int:
         recv_all = int(tot.item())
     else:
         recv_all = recv_total
+
+    if args.dump_steps and rank == 0:
+        print("step_times_ms:", [round(t * 1000, 3) for t in step_times])
 
     msgs_per_s = recv_all / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
