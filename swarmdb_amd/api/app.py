@@ -412,6 +412,27 @@ def create_app(
         )
         return [MessageResponse.from_message(m) for m in msgs]
 
+    @app.get("/messages/search/", response_model=List[MessageResponse])
+    async def search_messages(
+        keyword: str = Query(..., min_length=1, max_length=256),
+        case_sensitive: bool = Query(False),
+        limit: int = Query(100, ge=1, le=1000),
+        current: str = Depends(get_current_agent),
+    ):
+        """Content keyword search (search_messages was unrouted in the
+        reference — swarmdb/ main.py:742-781). Runs the device scan
+        kernel on GPU; non-admins see only their own traffic."""
+        msgs = db.search_messages(keyword, case_sensitive=case_sensitive,
+                                  limit=limit)
+        if not is_admin(current):
+            msgs = [
+                m for m in msgs
+                if m.sender_id == current
+                or m.receiver_id == current
+                or current in m.visible_to
+            ]
+        return [MessageResponse.from_message(m) for m in msgs]
+
     @app.get("/agents/{agent_id}/messages", response_model=List[MessageResponse])
     async def get_agent_messages(
         agent_id: str,

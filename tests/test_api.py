@@ -375,3 +375,20 @@ def test_batch_send_endpoint(client):
     # empty batch
     r = client.post("/messages/batch", headers=ha, json=[])
     assert r.json()["message_ids"] == []
+
+
+def test_search_endpoint(client):
+    ha, hb, hc = auth(client, "alice"), auth(client, "bob"), auth(client, "carol")
+    client.post("/messages", headers=ha,
+                json={"receiver_id": "bob", "content": "wombat sighting"})
+    client.post("/messages", headers=hc,
+                json={"receiver_id": "dave", "content": "wombat relocation"})
+    # non-admin sees only own traffic
+    r = client.get("/messages/search/?keyword=wombat", headers=hb)
+    assert [m["content"] for m in r.json()] == ["wombat sighting"]
+    # admin sees all
+    r = client.get("/messages/search/?keyword=wombat",
+                   headers=auth(client, "admin"))
+    assert len(r.json()) == 2
+    # no auth
+    assert client.get("/messages/search/?keyword=x").status_code == 401
