@@ -649,6 +649,8 @@ public:
     HIP_CHECK(hipEventCreateWithFlags(&ev_, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&up_ev_[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&up_ev_[1], hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&d2h_ev_[0], hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&d2h_ev_[1], hipEventDisableTiming));
 
     // device state
     HIP_CHECK(hipMalloc(&d_hdr_, (size_t)num_slots * sizeof(Rec)));
@@ -752,6 +754,8 @@ public:
     (void)hipEventDestroy(ev_);
     (void)hipEventDestroy(up_ev_[0]);
     (void)hipEventDestroy(up_ev_[1]);
+    (void)hipEventDestroy(d2h_ev_[0]);
+    (void)hipEventDestroy(d2h_ev_[1]);
     (void)hipStreamDestroy(stream_);
     (void)hipStreamDestroy(copy_stream_);
     (void)hipStreamDestroy(h2d_stream_);
@@ -1154,6 +1158,13 @@ public:
         for (int i = 0; i < n; ++i)
           bytes += h_fetch_hdr_[i].payload_len;
       } else {
+        // bounded async pipeline: wait for the delivery issued two
+        // calls ago before letting this one fly (an unbounded copy
+        // stream backlog eventually hard-blocks the runtime for the
+        // whole accumulated drain)
+        HIP_CHECK(hipEventRecord(d2h_ev_[d2h_cur_], copy_stream_));
+        d2h_cur_ ^= 1;
+        HIP_CHECK(hipEventSynchronize(d2h_ev_[d2h_cur_]));
         bytes = (u64)n * stride; // upper bound; D2H still in flight
       }
     }
@@ -1438,6 +1449,8 @@ private:
   hipStream_t stream_{}, copy_stream_{}, h2d_stream_{};
   hipEvent_t ev_{};
   hipEvent_t up_ev_[2] = {};
+  hipEvent_t d2h_ev_[2] = {};
+  int d2h_cur_ = 0;
   bool uploaded_[2] = {false, false};
 
   Rec *d_hdr_{};
