@@ -245,6 +245,64 @@ def config6_search(seconds: float) -> None:
     }))
 
 
+def config7_checkpoint(seconds: float) -> None:
+    """Bonus: checkpoint/restore throughput — JSON history spill of a
+    GPU-resident store (device->pinned gather on the copy stream + host
+    serialization) and the batched replay load."""
+    import shutil
+    import tempfile
+
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    tmp = tempfile.mkdtemp(prefix="swarmdb_ckpt_")
+    n_msgs = 1 << 17  # 128k x 256B messages
+    cfg = QueueConfig(use_gpu=_gpu_available(), auto_save=False,
+                      max_agents=256, num_slots=n_msgs,
+                      slot_bytes=512, staging_batch=1 << 14,
+                      save_dir=tmp)
+    db = SwarmsDB(config=cfg)
+    rng = np.random.default_rng(0)
+    agents = np.arange(64)
+    for a in agents:
+        db.agent_index(f"agent{a}")
+    idxs = np.array([db.agent_index(f"agent{a}") for a in agents])
+    batch = 1 << 14
+    for _ in range(n_msgs // batch):
+        recs, payload = _make_batch(rng, batch, idxs, idxs, 256)
+        db.send_batch(recs, payload)
+
+    t0 = time.perf_counter()
+    path = db.save_message_history()
+    save_s = time.perf_counter() - t0
+    size_mb = Path(path).stat().st_size / 1e6
+
+    cfg2 = QueueConfig(use_gpu=_gpu_available(), auto_save=False,
+                       max_agents=256, num_slots=n_msgs, slot_bytes=512,
+                       staging_batch=1 << 14, save_dir=tmp)
+    db2 = SwarmsDB(config=cfg2)
+    t0 = time.perf_counter()
+    loaded = db2.load_message_history(path)
+    load_s = time.perf_counter() - t0
+    assert loaded == n_msgs, (loaded, n_msgs)
+
+    db.config.auto_save = False
+    db2.config.auto_save = False
+    db.close()
+    db2.close()
+    shutil.rmtree(tmp, ignore_errors=True)
+    print(json.dumps({
+        "config": 7,
+        "name": "checkpoint-restore",
+        "messages": n_msgs,
+        "file_mb": round(size_mb, 1),
+        "save_s": round(save_s, 2),
+        "save_msgs_per_s": round(n_msgs / save_s, 0),
+        "load_s": round(load_s, 2),
+        "load_msgs_per_s": round(n_msgs / load_s, 0),
+        "engine": "gpu" if _gpu_available() else "cpu",
+    }))
+
+
 def config5_loadbalancer(seconds: float) -> None:
     """8 mock backends (1 per GPU on a full node); least-loaded dispatch
     via the wavefront min-reduce kernel at >=100k req/s, with concurrent
@@ -298,7 +356,7 @@ def config5_loadbalancer(seconds: float) -> None:
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--config", type=int, required=True,
-                    choices=[1, 2, 3, 4, 5, 6])
+                    choices=[1, 2, 3, 4, 5, 6, 7])
     ap.add_argument("--seconds", type=float, default=5.0)
     args = ap.parse_args()
     if args.config == 1:
@@ -323,6 +381,8 @@ def main() -> int:
         config5_loadbalancer(args.seconds)
     elif args.config == 6:
         config6_search(args.seconds)
+    elif args.config == 7:
+        config7_checkpoint(args.seconds)
     return 0
 
 

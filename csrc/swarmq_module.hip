@@ -73,25 +73,24 @@ static void par_memcpy(void *dst, const void *src, size_t n) {
 // kernels
 // ---------------------------------------------------------------------------
 
-// Enqueue a staged batch: one 64-lane wavefront per message. The wave
-// cooperatively copies the payload in 16-B chunks (64 lanes x 16 B = 1 KiB
-// per round — a 1 KiB chat message is ONE vector round); lane 0 writes the
-// header, status word (the DELIVERED ack, replacing the Kafka delivery
-// callback, reference "swarmdb/ main.py":374-391), counters, and the
-// point-to-point inbox append. Broadcast messages are queued into
-// bcast_list for k_fanout.
-__global__ void k_enqueue(const Rec *__restrict__ stage,
-                          const u8 *__restrict__ stage_pay, int n,
-                          u64 base_seq, Rec *__restrict__ hdr,
-                          u32 *__restrict__ status, u8 *__restrict__ payload,
-                          u64 *__restrict__ inbox, ull *__restrict__ inbox_wpos,
-                          ull *__restrict__ by_type,
-                          ull *__restrict__ by_status, ull *__restrict__ sent,
-                          u64 *__restrict__ bcast_list,
-                          u32 *__restrict__ bcast_count, QueueGeom g) {
-  // per-block type histogram: one global atomic per bin per block instead
-  // of one per message (single hot counter words otherwise serialize the
-  // whole batch through L2 atomics)
+// Enqueue, split in two for occupancy: k_enqueue_meta does the
+// per-message bookkeeping THREAD-per-message (header write, status ack,
+// counters, inbox append — 64 messages per wave instead of one message
+// per wave with 63 idle lanes), k_enqueue_payload streams the payload
+// bytes WAVE-per-message (64 lanes x 16 B = 1 KiB per vector round).
+// Both read the same staged records; they are independent and the
+// dequeue that needs both runs later on the same stream.
+__global__ void k_enqueue_meta(const Rec *__restrict__ stage, int n,
+                               u64 base_seq, Rec *__restrict__ hdr,
+                               u32 *__restrict__ status,
+                               u64 *__restrict__ inbox,
+                               ull *__restrict__ inbox_wpos,
+                               ull *__restrict__ by_type,
+                               ull *__restrict__ by_status,
+                               ull *__restrict__ sent,
+                               u64 *__restrict__ bcast_list,
+                               u32 *__restrict__ bcast_count, QueueGeom g) {
+  // per-block histograms: one global atomic per bin per block
   __shared__ u32 h_type[N_TYPES];
   __shared__ u32 h_msgs;
   __shared__ u32 h_failed;
@@ -103,16 +102,14 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
     h_failed = 0;
   __syncthreads();
 
-  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  const int lane = threadIdx.x & 63;
-  if (wave < n) {
-    const Rec r = stage[wave];
-    const u64 seq = base_seq + (u64)wave;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    const Rec r = stage[i];
+    const u64 seq = base_seq + (u64)i;
     const u32 slot = (u32)(seq % g.num_slots);
 
-    // the error lane (reference _errors topic analog, "swarmdb/
-    // main.py":260-273, 501-519): malformed records are parked FAILED
-    // — queryable via status — instead of corrupting device state
+    // error lane (reference _errors topic analog, "swarmdb/
+    // main.py":260-273, 501-519): malformed records park FAILED
     const bool bad =
         r.type >= N_TYPES || r.priority > 3 ||
         r.payload_len > g.slot_bytes ||
@@ -120,43 +117,27 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
         r.sender >= g.max_agents ||
         (r.vis_mode != VIS_ALL && r.bitmap != NO_BITMAP &&
          r.bitmap >= g.num_bitmaps);
+    Rec h = r;
+    h.payload_off = (u64)slot * g.slot_bytes;
     if (bad) {
-      if (lane == 0) {
-        Rec h = r;
-        h.payload_off = (u64)slot * g.slot_bytes;
-        h.payload_len = 0;
-        h.content_len = 0;
-        hdr[slot] = h;
-        status[slot] = ST_FAILED;
-        atomicAdd(&h_failed, 1u);
-      }
+      h.payload_len = 0;
+      h.content_len = 0;
+      hdr[slot] = h;
+      status[slot] = ST_FAILED;
+      atomicAdd(&h_failed, 1u);
     } else {
-      // payload copy, 16 B per lane per round (staging offsets are 16-B
-      // aligned; slots are slot_bytes-strided so destination is aligned)
-      const uint4 *src =
-          reinterpret_cast<const uint4 *>(stage_pay + r.payload_off);
-      uint4 *dst =
-          reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
-      const u32 nchunk = (r.payload_len + 15u) >> 4;
-      for (u32 c = lane; c < nchunk; c += 64)
-        dst[c] = src[c];
-
-      if (lane == 0) {
-        Rec h = r;
-        h.payload_off = (u64)slot * g.slot_bytes;
-        hdr[slot] = h;
-        status[slot] = ST_DELIVERED;
-        atomicAdd(&h_type[r.type], 1u);
-        atomicAdd(&h_msgs, 1u);
-        atomicAdd(&sent[r.sender], 1ull);
-        if (r.receiver == BROADCAST) {
-          u32 bi = atomicAdd(bcast_count, 1u);
-          bcast_list[bi] = seq;
-        } else {
-          ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
-          inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
-              seq;
-        }
+      hdr[slot] = h;
+      status[slot] = ST_DELIVERED;
+      atomicAdd(&h_type[r.type], 1u);
+      atomicAdd(&h_msgs, 1u);
+      atomicAdd(&sent[r.sender], 1ull);
+      if (r.receiver == BROADCAST) {
+        u32 bi = atomicAdd(bcast_count, 1u);
+        bcast_list[bi] = seq;
+      } else {
+        ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
+        inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
+            seq;
       }
     }
   }
@@ -167,6 +148,29 @@ __global__ void k_enqueue(const Rec *__restrict__ stage,
     atomicAdd(&by_status[ST_DELIVERED], (ull)h_msgs);
   if (threadIdx.x == N_TYPES + 1 && h_failed)
     atomicAdd(&by_status[ST_FAILED], (ull)h_failed);
+}
+
+__global__ void k_enqueue_payload(const Rec *__restrict__ stage,
+                                  const u8 *__restrict__ stage_pay, int n,
+                                  u64 base_seq, u8 *__restrict__ payload,
+                                  QueueGeom g) {
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (wave >= n)
+    return;
+  const u64 payload_off = stage[wave].payload_off;
+  const u32 payload_len = stage[wave].payload_len;
+  if (payload_len == 0 || payload_len > g.slot_bytes)
+    return;
+  const u64 seq = base_seq + (u64)wave;
+  const u32 slot = (u32)(seq % g.num_slots);
+  const uint4 *src =
+      reinterpret_cast<const uint4 *>(stage_pay + payload_off);
+  uint4 *dst =
+      reinterpret_cast<uint4 *>(payload + (u64)slot * g.slot_bytes);
+  const u32 nchunk = (payload_len + 15u) >> 4;
+  for (u32 c = lane; c < nchunk; c += 64)
+    dst[c] = src[c];
 }
 
 // Broadcast fan-out: one thread per agent appends the batch's broadcast
@@ -813,12 +817,13 @@ public:
                                  hipMemcpyHostToDevice, stream_));
       HIP_CHECK(hipEventRecord(stage_ev_[0], stream_));
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      const int waves_per_block = 4; // 256 threads
-      const int blocks = (n + waves_per_block - 1) / waves_per_block;
-      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
-                         d_stage_recs_[0], d_stage_pay_[0], n, base, d_hdr_,
-                         d_status_, d_payload_, d_inbox_, d_wpos_, d_by_type_,
+      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
+                         stream_, d_stage_recs_[0], n, base, d_hdr_,
+                         d_status_, d_inbox_, d_wpos_, d_by_type_,
                          d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
+                         stream_, d_stage_recs_[0], d_stage_pay_[0], n, base,
+                         d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
                          0, stream_, d_bcast_, d_bcast_count_, d_active_,
                          d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
@@ -956,12 +961,13 @@ public:
         HIP_CHECK(hipEventRecord(stage_ev_[slot], stream_));
       }
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      const int blocks = (n + 3) / 4;
-      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
-                         d_stage_recs_[slot], d_stage_pay_[slot], n, base,
-                         d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
-                         d_by_type_, d_by_status_, d_sent_, d_bcast_,
-                         d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
+                         stream_, d_stage_recs_[slot], n, base, d_hdr_,
+                         d_status_, d_inbox_, d_wpos_, d_by_type_,
+                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
+                         stream_, d_stage_recs_[slot], d_stage_pay_[slot], n,
+                         base, d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
                          dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
                          d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
@@ -1021,13 +1027,15 @@ public:
     {
       py::gil_scoped_release nogil;
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      const int blocks = (n + 3) / 4;
-      hipLaunchKernelGGL(k_enqueue, dim3(blocks), dim3(256), 0, stream_,
-                         reinterpret_cast<const Rec *>(recs_ptr),
-                         reinterpret_cast<const u8 *>(pay_ptr), n, base,
-                         d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
+      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
+                         stream_, reinterpret_cast<const Rec *>(recs_ptr), n,
+                         base, d_hdr_, d_status_, d_inbox_, d_wpos_,
                          d_by_type_, d_by_status_, d_sent_, d_bcast_,
                          d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
+                         stream_, reinterpret_cast<const Rec *>(recs_ptr),
+                         reinterpret_cast<const u8 *>(pay_ptr), n, base,
+                         d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
                          dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
                          d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
