@@ -106,8 +106,10 @@ class Message(BaseModel):
         }
 
     @classmethod
-    def from_dict(cls, data: Dict[str, Any]) -> "Message":
-        """Inverse of :meth:`to_dict` (reference swarmdb/ main.py:101-111)."""
+    def from_dict(cls, data: Dict[str, Any], validate: bool = True) -> "Message":
+        """Inverse of :meth:`to_dict` (reference swarmdb/ main.py:101-111).
+        ``validate=False`` skips pydantic validation for trusted input
+        (our own history files on the bulk load path)."""
         d = dict(data)
         if "type" in d:
             d["type"] = MessageType(d["type"])
@@ -115,4 +117,10 @@ class Message(BaseModel):
             d["priority"] = MessagePriority(d["priority"])
         if "status" in d:
             d["status"] = MessageStatus(d["status"])
+        if not validate:
+            d.setdefault("metadata", {})
+            d.setdefault("visible_to", [])
+            d.setdefault("token_count", None)
+            d.setdefault("receiver_id", None)
+            return cls.model_construct(**d)
         return cls(**d)

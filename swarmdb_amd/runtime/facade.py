@@ -941,7 +941,10 @@ class SwarmsDB:
             for agent_id in history.get("registered_agents", []):
                 self.register_agent(agent_id)
             msgs = sorted(
-                (Message.from_dict(d) for d in history.get("messages", {}).values()),
+                (
+                    Message.from_dict(d, validate=False)
+                    for d in history.get("messages", {}).values()
+                ),
                 key=lambda m: m.timestamp,
             )
             for m in msgs:
