@@ -11,8 +11,9 @@ RCCL over xGMI on GPU, gloo on CPU for CI). Replaces the reference's
   dense agent-index table and bitmap pool stay bit-identical across
   ranks with no coordinator;
 - **data plane**: outbound messages batch into a per-tick exchange
-  routed by owner rank (``shard_for`` over the stable hash) via
-  all-to-all; an agent's inbox lives only on its owner rank.
+  routed by owner rank (dense agent index mod world — the same mapping
+  the all-to-all router applies) via all-to-all; an agent's inbox lives
+  only on its owner rank.
 
 Usage::
 
