@@ -69,6 +69,9 @@ VIS_GROUP = 2    # group fan-out: inbox entries land ONLY in member inboxes
 FLAG_HAS_EXTRAS = 1   # payload tail carries an extras-JSON blob
 FLAG_JSON_CONTENT = 2  # content bytes are JSON (dict/list), else raw utf-8 str
 FLAG_DERIVED_ID = 4   # message id is derived from (rank, seq), no extras id
+FLAG_OVERFLOW = 8     # payload exceeds the slot capacity and lives in the
+#                       facade's host-side overflow store (device slot empty;
+#                       not visible to the device search kernel)
 
 # per-message batch record (host staging). 48 bytes, 8-byte aligned.
 # Payload layout: [content bytes (content_len)][extras JSON (rest)] —

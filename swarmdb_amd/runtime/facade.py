@@ -49,6 +49,7 @@ from .engine import (
     BROADCAST,
     FLAG_HAS_EXTRAS,
     FLAG_JSON_CONTENT,
+    FLAG_OVERFLOW,
     NO_BITMAP,
     REC_DTYPE,
     ST_DELETED,
@@ -132,6 +133,10 @@ class SwarmsDB:
         self.metadata: Dict[str, Any] = {}       # groups + llm assignments
         self._id_to_seq: Dict[str, int] = {}     # compat-path id map
         self._failed: Dict[str, Message] = {}    # failed sends for resend
+        # host-side store for payloads larger than a device slot:
+        # seq -> (content_len, payload bytes)
+        # (SURVEY.md §7 'variable-size content in fixed slots')
+        self._overflow: Dict[int, tuple] = {}
 
         self._llm_backends: List[str] = []
         self._llm_backend_idx: Dict[str, int] = {}
@@ -227,6 +232,9 @@ class SwarmsDB:
         made deterministic per SURVEY.md §8.6)."""
         return partition_for(agent_id, self.config.num_partitions)
 
+    def _slot_capacity(self) -> int:
+        return int(self.config.slot_bytes)
+
     def _bitmap_for(self, visible_to: List[str]) -> int:
         bits = np.zeros(self.config.max_agents, dtype=bool)
         for a in visible_to:
@@ -283,6 +291,21 @@ class SwarmsDB:
             content_b, is_json = encode_content(content)
             extras_b = encode_extras(msg.id, msg.metadata, vis)
             payload = content_b + extras_b
+            flags = FLAG_HAS_EXTRAS
+            if is_json:
+                flags |= FLAG_JSON_CONTENT
+            overflow_payload = None
+            if len(payload) > self._slot_capacity():
+                # oversized content: the device slot stays empty and the
+                # payload lives host-side; routing/delivery is unchanged
+                overflow_payload = payload
+                payload = b""
+                flags |= FLAG_OVERFLOW
+                rec["payload_len"] = 0
+                rec["content_len"] = 0
+            else:
+                rec["payload_len"] = len(payload)
+                rec["content_len"] = len(content_b)
             rec["sender"] = self._agent_idx[sender_id]
             rec["receiver"] = (
                 BROADCAST if receiver_id is None else self._agent_idx[receiver_id]
@@ -292,11 +315,6 @@ class SwarmsDB:
             rec["timestamp"] = msg.timestamp
             rec["token_count"] = token_count
             rec["payload_off"] = 0
-            rec["payload_len"] = len(payload)
-            rec["content_len"] = len(content_b)
-            flags = FLAG_HAS_EXTRAS
-            if is_json:
-                flags |= FLAG_JSON_CONTENT
             rec["flags"] = flags
             if vis:
                 rec["vis_mode"] = VIS_BITMAP
@@ -316,6 +334,8 @@ class SwarmsDB:
                 logger.error("send failed for %s: %s", msg.id, e)
                 raise
             self._id_to_seq[msg.id] = int(seqs[0])
+            if overflow_payload is not None:
+                self._overflow[int(seqs[0])] = (len(content_b), overflow_payload)
             self._maybe_autosave()
             return msg.id
 
@@ -525,8 +545,11 @@ class SwarmsDB:
         return out
 
     def _reconstruct(self, row: np.void, payload: bytes) -> Message:
-        clen = int(row["content_len"])
         flags = int(row["flags"])
+        if flags & FLAG_OVERFLOW:
+            clen, payload = self._overflow.get(int(row["seq"]), (0, b""))
+        else:
+            clen = int(row["content_len"])
         content = decode_content(payload[:clen], bool(flags & FLAG_JSON_CONTENT))
         extras = (
             decode_extras(payload[clen:]) if flags & FLAG_HAS_EXTRAS else {}
@@ -816,8 +839,11 @@ class SwarmsDB:
     def _message_dict(self, row: np.void, payload: bytes) -> Dict[str, Any]:
         """Reconstruct the wire dict straight from an engine row —
         the spill path skips pydantic construction entirely."""
-        clen = int(row["content_len"])
         flags = int(row["flags"])
+        if flags & FLAG_OVERFLOW:
+            clen, payload = self._overflow.get(int(row["seq"]), (0, b""))
+        else:
+            clen = int(row["content_len"])
         extras = (
             decode_extras(payload[clen:]) if flags & FLAG_HAS_EXTRAS else {}
         )
@@ -957,10 +983,16 @@ class SwarmsDB:
             recs = np.zeros(n, dtype=REC_DTYPE)
             chunks: List[bytes] = []
             off = 0
+            overflow_pending: Dict[int, tuple] = {}
             for i, m in enumerate(msgs):
                 content_b, is_json = encode_content(m.content)
                 extras_b = encode_extras(m.id, m.metadata, m.visible_to)
                 payload = content_b + extras_b
+                if len(payload) > self._slot_capacity():
+                    overflow_pending[i] = (len(content_b), payload)
+                    recs["flags"][i] = FLAG_OVERFLOW
+                    payload = b""
+                    content_b = b""
                 pad = (-len(payload)) % 16
                 chunks.append(payload + b"\x00" * pad)
                 recs["sender"][i] = self._agent_idx[m.sender_id]
@@ -976,7 +1008,7 @@ class SwarmsDB:
                 recs["payload_off"][i] = off
                 recs["payload_len"][i] = len(payload)
                 recs["content_len"][i] = len(content_b)
-                recs["flags"][i] = FLAG_HAS_EXTRAS | (
+                recs["flags"][i] |= FLAG_HAS_EXTRAS | (
                     FLAG_JSON_CONTENT if is_json else 0
                 )
                 if m.visible_to:
@@ -990,6 +1022,8 @@ class SwarmsDB:
                 seqs = self.engine.enqueue_batch(recs, b"".join(chunks))
                 for m, s in zip(msgs, seqs):
                     self._id_to_seq[m.id] = int(s)
+                for i, blob in overflow_pending.items():
+                    self._overflow[int(seqs[i])] = blob
                 self.engine.set_statuses(
                     seqs,
                     np.fromiter(

@@ -407,3 +407,26 @@ def test_error_lane_malformed_batch(tmp_db):
     assert set(failed.tolist()) == set(seqs[1:].tolist())
     s = tmp_db.get_stats()
     assert s["messages_by_status"].get("failed") == 3
+
+
+def test_oversized_content_overflow_store(tmp_db):
+    """Content larger than a device slot routes normally; the payload
+    lives in the host-side overflow store (SURVEY.md §7 hard part 2)."""
+    big = "Z" * (tmp_db.config.slot_bytes * 3)
+    mid = tmp_db.send_message("a", big, receiver_id="b",
+                              metadata={"note": "huge"})
+    got = tmp_db.receive_messages("b", timeout=0)
+    assert len(got) == 1
+    assert got[0].id == mid
+    assert got[0].content == big
+    assert got[0].metadata == {"note": "huge"}
+    # round-trips through get_message and history save
+    assert tmp_db.get_message(mid).content == big
+    path = tmp_db.save_message_history()
+    data = json.loads(open(path).read())
+    assert data["messages"][mid]["content"] == big
+    # dict/list content too
+    big_dict = {"payload": ["x" * 1000] * 10}
+    mid2 = tmp_db.send_message("a", big_dict, receiver_id="b")
+    assert tmp_db.receive_messages("b", timeout=0)[0].content == big_dict
+    assert tmp_db.get_message(mid2).content == big_dict

@@ -134,3 +134,18 @@ def test_groups_sidecar_persisted(tmp_path):
     assert side["llm_backends"] == {"a": "backendX"}
     db.config.auto_save = False
     db.close()
+
+
+def test_oversized_content_survives_save_load(tmp_path):
+    db = _mk(tmp_path)
+    big = "Q" * (db.config.slot_bytes * 2)
+    mid = db.send_message("a", big, receiver_id="b")
+    path = db.save_message_history()
+    db.config.auto_save = False
+    db.close()
+    db2 = _mk(tmp_path)
+    db2.load_message_history(path)
+    assert db2.get_message(mid).content == big
+    assert db2.receive_messages("b", timeout=0)[0].content == big
+    db2.config.auto_save = False
+    db2.close()
