@@ -732,6 +732,7 @@ class SwarmsDB:
                 return False
             ok = self.engine.delete(seq)
             self._id_to_seq.pop(message_id, None)
+            self._overflow.pop(seq, None)
             return ok
 
     # ------------------------------------------------------------------
@@ -1065,6 +1066,7 @@ class SwarmsDB:
                     if self.engine.delete(int(s)):
                         archive[m.id] = m.to_dict()
                         self._id_to_seq.pop(m.id, None)
+                        self._overflow.pop(int(s), None)
                         count += 1
         if count == 0:
             return 0
