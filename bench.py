@@ -170,10 +170,30 @@ def main() -> int:
             torch.cuda.synchronize()
 
     def step(i: int) -> int:
-        nonlocal sent_total, recv_total
+        nonlocal sent_total, recv_total, grouter, router
         recs, payload = batches[i % len(batches)]
         if grouter is not None:
-            sent_local = grouter.route_and_enqueue(recs, payload)
+            try:
+                sent_local = grouter.route_and_enqueue(recs, payload)
+            except Exception as e:
+                # first-line fallback: host-path router still uses RCCL
+                # for the exchange but routes via host buffers — the
+                # scaling run survives a GPU-direct path failure
+                if i == 0:
+                    print(f"[bench] GpuDirectRouter failed ({e}); "
+                          "falling back to host-path router",
+                          file=sys.stderr)
+                    from swarmdb_amd.parallel.router import CrossGpuRouter
+
+                    router = CrossGpuRouter(
+                        torch.device("cuda", local_rank)
+                    )
+                    grouter = None
+                    recs2, payload2 = router.route(recs, payload)
+                    engine.enqueue_batch(recs2, payload2)
+                    sent_local = len(recs2)
+                else:
+                    raise
         else:
             if router is not None:
                 recs, payload = router.route(recs, payload)
