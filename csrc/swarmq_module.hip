@@ -1434,7 +1434,10 @@ private:
     size_t nb = stage_pay_bytes_;
     while (nb < bytes)
       nb *= 2;
+    // all three streams may hold work against the staging buffers
     HIP_CHECK(hipStreamSynchronize(stream_));
+    HIP_CHECK(hipStreamSynchronize(h2d_stream_));
+    HIP_CHECK(hipStreamSynchronize(copy_stream_));
     for (int s = 0; s < 2; ++s) {
       HIP_CHECK(hipFree(d_stage_pay_[s]));
       (void)hipHostFree(h_pay_[s]);
