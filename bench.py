@@ -107,7 +107,15 @@ def main() -> int:
 
         if have_gpu:
             torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
-        dist.init_process_group(backend=backend)
+        if backend == "nccl" and have_gpu:
+            dist.init_process_group(
+                backend=backend,
+                device_id=torch.device(
+                    "cuda", local_rank % max(1, torch.cuda.device_count())
+                ),
+            )
+        else:
+            dist.init_process_group(backend=backend)
 
     agents_global = args.agents * world
     max_agents = ((agents_global + 63) // 64) * 64
