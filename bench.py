@@ -89,6 +89,8 @@ def main() -> int:
                     help="skip payload D2H gather (delivery stays device-side)")
     ap.add_argument("--dump-steps", action="store_true",
                     help="print per-step wall times (variance diagnosis)")
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable the hipGraph-captured tick")
     args = ap.parse_args()
 
     import torch
@@ -251,18 +253,43 @@ def main() -> int:
 
         _prefetch(0, 0)
 
+        # hipGraph-captured steady-state tick: enqueue + fanout +
+        # receive + result D2H replay as ONE graph launch per step
+        use_graph = not args.no_graph
+        if use_graph:
+            try:
+                q.build_tick(args.batch, local_agents.astype(np.uint32),
+                             recv_K, bool(args.priority))
+            except Exception as e:
+                print(f"[bench] tick-graph capture unavailable ({e}); "
+                      "using discrete launches", file=sys.stderr)
+                use_graph = False
+
+        kidx = np.arange(recv_K)[None, :]
+
         def step(i: int, _cur=[0]) -> int:  # noqa: F811
             nonlocal sent_total, recv_total
             cur = _cur[0]
             n_staged = len(batches[i % len(batches)][0])
-            q.enqueue_staged(cur)
-            # upload batch i+1 on the H2D stream: overlaps batch i's
-            # kernels and the delivery D2H (full-duplex PCIe)
-            _prefetch(1 - cur, i + 1)
-            counts, seqs = engine.receive_many(
-                local_agents, recv_K, priority_order=args.priority
-            )
-            ndel = int(counts.sum())
+            if use_graph:
+                # upload batch i+1 first so its H2D overlaps this tick's
+                # graph execution
+                _prefetch(1 - cur, i + 1)
+                counts, flat = q.run_tick(cur)
+                counts = counts.astype(np.int64)
+                ndel = int(counts.sum())
+                seqs = flat.reshape(len(local_agents), recv_K)[
+                    kidx < counts[:, None]
+                ]
+            else:
+                q.enqueue_staged(cur)
+                # upload batch i+1 on the H2D stream: overlaps batch i's
+                # kernels and the delivery D2H (full-duplex PCIe)
+                _prefetch(1 - cur, i + 1)
+                counts, seqs = engine.receive_many(
+                    local_agents, recv_K, priority_order=args.priority
+                )
+                ndel = int(counts.sum())
             if not args.no_gather and ndel:
                 # async: the D2H overlaps the next tick's H2D staging
                 engine.deliver_payloads(seqs, args.payload, synchronize=False)

@@ -80,8 +80,22 @@ static void par_memcpy(void *dst, const void *src, size_t n) {
 // bytes WAVE-per-message (64 lanes x 16 B = 1 KiB per vector round).
 // Both read the same staged records; they are independent and the
 // dequeue that needs both runs later on the same stream.
+// Graph-tick prologue: advance the device-side tail and publish this
+// tick's (base, evict) pair for the captured kernels to read.
+__global__ void k_tick_begin(u64 *__restrict__ base_evict,
+                             ull *__restrict__ tail, int n, u64 num_slots) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    const u64 base = *tail;
+    base_evict[0] = base;
+    const u64 cnt = base + (u64)n;
+    base_evict[1] = cnt > num_slots ? cnt - num_slots : 0;
+    *tail = cnt;
+  }
+}
+
 __global__ void k_enqueue_meta(const Rec *__restrict__ stage, int n,
-                               u64 base_seq, Rec *__restrict__ hdr,
+                               u64 base_seq, const u64 *__restrict__ dyn,
+                               Rec *__restrict__ hdr,
                                u32 *__restrict__ status,
                                u64 *__restrict__ inbox,
                                ull *__restrict__ inbox_wpos,
@@ -105,7 +119,7 @@ __global__ void k_enqueue_meta(const Rec *__restrict__ stage, int n,
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) {
     const Rec r = stage[i];
-    const u64 seq = base_seq + (u64)i;
+    const u64 seq = (dyn ? dyn[0] : base_seq) + (u64)i;
     const u32 slot = (u32)(seq % g.num_slots);
 
     // error lane (reference _errors topic analog, "swarmdb/
@@ -152,8 +166,8 @@ __global__ void k_enqueue_meta(const Rec *__restrict__ stage, int n,
 
 __global__ void k_enqueue_payload(const Rec *__restrict__ stage,
                                   const u8 *__restrict__ stage_pay, int n,
-                                  u64 base_seq, u8 *__restrict__ payload,
-                                  QueueGeom g) {
+                                  u64 base_seq, const u64 *__restrict__ dyn,
+                                  u8 *__restrict__ payload, QueueGeom g) {
   const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   if (wave >= n)
@@ -162,7 +176,7 @@ __global__ void k_enqueue_payload(const Rec *__restrict__ stage,
   const u32 payload_len = stage[wave].payload_len;
   if (payload_len == 0 || payload_len > g.slot_bytes)
     return;
-  const u64 seq = base_seq + (u64)wave;
+  const u64 seq = (dyn ? dyn[0] : base_seq) + (u64)wave;
   const u32 slot = (u32)(seq % g.num_slots);
   const uint4 *src =
       reinterpret_cast<const uint4 *>(stage_pay + payload_off);
@@ -218,7 +232,8 @@ __global__ void k_fanout(const u64 *__restrict__ bcast_list,
 // prefix READ, and carries leftovers.
 __global__ void __launch_bounds__(256)
     k_receive(const u32 *__restrict__ agents, int n_agents, int max_per_agent,
-              int priority_mode, u64 evict_base, const Rec *__restrict__ hdr,
+              int priority_mode, u64 evict_base,
+              const u64 *__restrict__ dyn, const Rec *__restrict__ hdr,
               u32 *__restrict__ status, const u64 *__restrict__ inbox,
               ull *__restrict__ inbox_wpos, ull *__restrict__ inbox_rpos,
               u64 *__restrict__ carry, u32 *__restrict__ carry_n,
@@ -231,6 +246,8 @@ __global__ void __launch_bounds__(256)
   const int b = blockIdx.x;
   if (b >= n_agents)
     return;
+  if (dyn)
+    evict_base = dyn[1];
   const u32 a = agents[b];
   const ull r = inbox_rpos[a];
   const ull w = inbox_wpos[a];
@@ -684,6 +701,9 @@ public:
     HIP_CHECK(hipMalloc(&d_agents_, (size_t)max_agents * sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_unread_, (size_t)max_agents * sizeof(u32)));
 
+    HIP_CHECK(hipMalloc(&d_dyn_, 2 * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_tail_, sizeof(ull)));
+    HIP_CHECK(hipMemset(d_tail_, 0, sizeof(ull)));
     out_pool_ = (size_t)4 << 20; // receive output pool: 4M entries, 32 MB
     HIP_CHECK(hipMalloc(&d_out_seqs_, out_pool_ * sizeof(u64)));
     HIP_CHECK(hipMalloc(&d_out_counts_, (size_t)max_agents * sizeof(u32)));
@@ -742,6 +762,7 @@ public:
                              d_received_, d_bcast_, d_bcast_count_,
                              d_backend_loads_, d_choices_, d_match_,
                              d_match_count_, d_needle_, d_agents_, d_unread_,
+                             d_dyn_, d_tail_,
                              d_out_seqs_, d_out_counts_, d_stage_recs_[0],
                              d_stage_recs_[1], d_stage_pay_[0],
                              d_stage_pay_[1], d_seqs_in_, d_fetch_hdr_,
@@ -755,6 +776,9 @@ public:
       (void)hipHostFree(p);
     (void)hipEventDestroy(stage_ev_[0]);
     (void)hipEventDestroy(stage_ev_[1]);
+    for (int s = 0; s < 2; ++s)
+      if (tick_exec_[s])
+        (void)hipGraphExecDestroy(tick_exec_[s]);
     (void)hipEventDestroy(ev_);
     (void)hipEventDestroy(up_ev_[0]);
     (void)hipEventDestroy(up_ev_[1]);
@@ -818,12 +842,12 @@ public:
       HIP_CHECK(hipEventRecord(stage_ev_[0], stream_));
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
       hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
-                         stream_, d_stage_recs_[0], n, base, d_hdr_,
-                         d_status_, d_inbox_, d_wpos_, d_by_type_,
+                         stream_, d_stage_recs_[0], n, base, (const u64 *)nullptr,
+                         d_hdr_, d_status_, d_inbox_, d_wpos_, d_by_type_,
                          d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
       hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
                          stream_, d_stage_recs_[0], d_stage_pay_[0], n, base,
-                         d_payload_, g_);
+                         (const u64 *)nullptr, d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
                          0, stream_, d_bcast_, d_bcast_count_, d_active_,
                          d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
@@ -962,12 +986,13 @@ public:
       }
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
       hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
-                         stream_, d_stage_recs_[slot], n, base, d_hdr_,
-                         d_status_, d_inbox_, d_wpos_, d_by_type_,
-                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+                         stream_, d_stage_recs_[slot], n, base,
+                         (const u64 *)nullptr, d_hdr_, d_status_, d_inbox_,
+                         d_wpos_, d_by_type_, d_by_status_, d_sent_, d_bcast_,
+                         d_bcast_count_, g_);
       hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
                          stream_, d_stage_recs_[slot], d_stage_pay_[slot], n,
-                         base, d_payload_, g_);
+                         base, (const u64 *)nullptr, d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
                          dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
                          d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
@@ -977,6 +1002,90 @@ public:
     if (count_ > g_.num_slots)
       evict_base_ = count_ - g_.num_slots;
     return base;
+  }
+
+  // ---- captured steady-state tick (hipGraph) ----
+  // One graph per staging slot captures the whole delivery tick:
+  // tick-begin (device tail) -> enqueue meta+payload -> fanout ->
+  // receive -> counts/seqs D2H. Replay cost is one graph launch instead
+  // of ~10 API calls; the H2D prefetch stays outside (its event is
+  // waited on the stream before the launch).
+  void build_tick(int n, py::array_t<u32> agents, int K, bool priority) {
+    const int na = (int)agents.size();
+    if (n <= 0 || (u32)n > staging_batch_)
+      throw std::invalid_argument("bad tick batch size");
+    if ((size_t)na * K > out_pool_)
+      throw std::invalid_argument("tick exceeds output pool");
+    for (int s = 0; s < 2; ++s)
+      if (tick_exec_[s]) {
+        (void)hipGraphExecDestroy(tick_exec_[s]);
+        tick_exec_[s] = nullptr;
+      }
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    HIP_CHECK(hipMemcpy(d_agents_, agents.data(), na * sizeof(u32),
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(d_tail_, &count_, sizeof(ull),
+                        hipMemcpyHostToDevice));
+    for (int s = 0; s < 2; ++s) {
+      hipGraph_t graph;
+      HIP_CHECK(hipStreamBeginCapture(stream_, hipStreamCaptureModeThreadLocal));
+      HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
+      hipLaunchKernelGGL(k_tick_begin, dim3(1), dim3(64), 0, stream_, d_dyn_,
+                         d_tail_, n, (u64)g_.num_slots);
+      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
+                         stream_, d_stage_recs_[s], n, 0, d_dyn_, d_hdr_,
+                         d_status_, d_inbox_, d_wpos_, d_by_type_,
+                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
+                         stream_, d_stage_recs_[s], d_stage_pay_[s], n, 0,
+                         d_dyn_, d_payload_, g_);
+      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
+                         dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
+                         d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
+                         g_);
+      hipLaunchKernelGGL(k_receive, dim3(na), dim3(256), 0, stream_,
+                         d_agents_, na, K, priority ? 1 : 0, 0, d_dyn_,
+                         d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
+                         d_carry_, d_carry_n_, d_bitmaps_, d_out_seqs_,
+                         d_out_counts_, d_by_status_, d_received_, g_);
+      HIP_CHECK(hipMemcpyAsync(h_out_counts_, d_out_counts_, na * sizeof(u32),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipMemcpyAsync(h_out_seqs_, d_out_seqs_,
+                               (size_t)na * K * sizeof(u64),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamEndCapture(stream_, &graph));
+      HIP_CHECK(hipGraphInstantiate(&tick_exec_[s], graph, nullptr, nullptr, 0));
+      HIP_CHECK(hipGraphDestroy(graph));
+    }
+    tick_n_ = n;
+    tick_na_ = na;
+    tick_K_ = K;
+  }
+
+  // Replay the captured tick for a slot previously uploaded with
+  // prefetch_from/prefetch_staged. Returns (counts, seqs).
+  py::tuple run_tick(int slot) {
+    if (slot < 0 || slot > 1 || !tick_exec_[slot])
+      throw std::invalid_argument("tick graph not built for this slot");
+    py::array_t<u32> counts(tick_na_);
+    py::array_t<u64> seqs((size_t)tick_na_ * tick_K_);
+    {
+      py::gil_scoped_release nogil;
+      if (uploaded_[slot]) {
+        HIP_CHECK(hipStreamWaitEvent(stream_, up_ev_[slot], 0));
+        uploaded_[slot] = false;
+      }
+      HIP_CHECK(hipGraphLaunch(tick_exec_[slot], stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    count_ += (u64)tick_n_;
+    if (count_ > g_.num_slots)
+      evict_base_ = count_ - g_.num_slots;
+    std::memcpy(counts.mutable_data(), h_out_counts_,
+                tick_na_ * sizeof(u32));
+    std::memcpy(seqs.mutable_data(), h_out_seqs_,
+                (size_t)tick_na_ * tick_K_ * sizeof(u64));
+    return py::make_tuple(counts, seqs);
   }
 
   // ---- GPU-direct cross-GPU routing support ----
@@ -1029,13 +1138,13 @@ public:
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
       hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
                          stream_, reinterpret_cast<const Rec *>(recs_ptr), n,
-                         base, d_hdr_, d_status_, d_inbox_, d_wpos_,
-                         d_by_type_, d_by_status_, d_sent_, d_bcast_,
-                         d_bcast_count_, g_);
+                         base, (const u64 *)nullptr, d_hdr_, d_status_,
+                         d_inbox_, d_wpos_, d_by_type_, d_by_status_, d_sent_,
+                         d_bcast_, d_bcast_count_, g_);
       hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
                          stream_, reinterpret_cast<const Rec *>(recs_ptr),
                          reinterpret_cast<const u8 *>(pay_ptr), n, base,
-                         d_payload_, g_);
+                         (const u64 *)nullptr, d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
                          dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
                          d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
@@ -1076,6 +1185,7 @@ public:
                                hipMemcpyHostToDevice, stream_));
       hipLaunchKernelGGL(k_receive, dim3(na), dim3(256), 0, stream_, d_agents_,
                          na, max_per_agent, priority ? 1 : 0, evict_base_,
+                         (const u64 *)nullptr,
                          d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
                          d_carry_, d_carry_n_, d_bitmaps_, d_out_seqs_,
                          d_out_counts_, d_by_status_, d_received_, g_);
@@ -1463,6 +1573,11 @@ private:
   hipEvent_t d2h_ev_[2] = {};
   int d2h_cur_ = 0;
   bool uploaded_[2] = {false, false};
+  // captured steady-state tick (one exec per staging slot)
+  hipGraphExec_t tick_exec_[2] = {};
+  int tick_n_ = 0, tick_na_ = 0, tick_K_ = 0;
+  u64 *d_dyn_{};   // [0]=tick base seq, [1]=evict base
+  ull *d_tail_{};  // device-side total-messages counter
 
   Rec *d_hdr_{};
   u32 *d_status_{};
@@ -1540,6 +1655,8 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("enqueue_staged", &DeviceQueue::enqueue_staged)
       .def("prefetch_staged", &DeviceQueue::prefetch_staged)
       .def("prefetch_from", &DeviceQueue::prefetch_from)
+      .def("build_tick", &DeviceQueue::build_tick)
+      .def("run_tick", &DeviceQueue::run_tick)
       .def_static("alloc_pinned", &DeviceQueue::alloc_pinned)
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
       .def("pack_exchange", &DeviceQueue::pack_exchange)
