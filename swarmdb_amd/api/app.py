@@ -101,6 +101,7 @@ def create_app(
     db: Optional[SwarmsDB] = None,
     settings: Optional[ApiSettings] = None,
     credential_validator: Optional[Callable[[str, str], bool]] = None,
+    batch_window: Optional[float] = None,
 ) -> FastAPI:
     """App factory. ``db`` defaults to a SwarmsDB built from env config
     (GPU engine if a device is visible, CPU otherwise). One shared
@@ -109,12 +110,27 @@ def create_app(
     settings = settings or ApiSettings()
     if db is None:
         db = SwarmsDB(config=QueueConfig.from_env())
+    if batch_window is None:
+        batch_window = float(os.environ.get("SWARMDB_BATCH_WINDOW", "0"))
+
+    # request micro-batcher: concurrent sends share one engine batch
+    # (SURVEY.md §7 hard part 3); off by default for exact per-request
+    # semantics
+    batcher = None
+    if batch_window > 0:
+        from .batcher import SendBatcher
+
+        batcher = SendBatcher(db, window=batch_window)
 
     from contextlib import asynccontextmanager
 
     @asynccontextmanager
     async def lifespan(app_: FastAPI):
+        if batcher is not None:
+            await batcher.start()
         yield
+        if batcher is not None:
+            await batcher.stop()
         # reference api.py:939-945 (shutdown hook)
         db.close()
 
@@ -260,17 +276,31 @@ def create_app(
     async def send_message(
         req: MessageRequest, current: str = Depends(get_current_agent)
     ):
-        """reference api.py:472-504 — sender is the authenticated agent."""
+        """reference api.py:472-504 — sender is the authenticated agent.
+        With a batch window configured, concurrent sends coalesce into
+        one engine batch via the micro-batcher."""
         try:
-            mid = db.send_message(
-                sender_id=current,
-                content=req.content,
-                receiver_id=req.receiver_id,
-                message_type=req.message_type,
-                priority=req.priority,
-                metadata=req.metadata,
-                visible_to=req.visible_to,
-            )
+            if batcher is not None:
+                msg = db.make_message(
+                    sender_id=current,
+                    content=req.content,
+                    receiver_id=req.receiver_id,
+                    message_type=req.message_type,
+                    priority=req.priority,
+                    metadata=req.metadata,
+                    visible_to=req.visible_to,
+                )
+                mid = await batcher.send(msg)
+            else:
+                mid = db.send_message(
+                    sender_id=current,
+                    content=req.content,
+                    receiver_id=req.receiver_id,
+                    message_type=req.message_type,
+                    priority=req.priority,
+                    metadata=req.metadata,
+                    visible_to=req.visible_to,
+                )
         except Exception as e:
             raise HTTPException(
                 status_code=http.HTTP_500_INTERNAL_SERVER_ERROR,

@@ -287,41 +287,7 @@ class SwarmsDB:
                 visible_to=vis,
             )
 
-            rec = np.zeros(1, dtype=REC_DTYPE)
-            content_b, is_json = encode_content(content)
-            extras_b = encode_extras(msg.id, msg.metadata, vis)
-            payload = content_b + extras_b
-            flags = FLAG_HAS_EXTRAS
-            if is_json:
-                flags |= FLAG_JSON_CONTENT
-            overflow_payload = None
-            if len(payload) > self._slot_capacity():
-                # oversized content: the device slot stays empty and the
-                # payload lives host-side; routing/delivery is unchanged
-                overflow_payload = payload
-                payload = b""
-                flags |= FLAG_OVERFLOW
-                rec["payload_len"] = 0
-                rec["content_len"] = 0
-            else:
-                rec["payload_len"] = len(payload)
-                rec["content_len"] = len(content_b)
-            rec["sender"] = self._agent_idx[sender_id]
-            rec["receiver"] = (
-                BROADCAST if receiver_id is None else self._agent_idx[receiver_id]
-            )
-            rec["type"] = _type_code(mtype)
-            rec["priority"] = prio.value
-            rec["timestamp"] = msg.timestamp
-            rec["token_count"] = token_count
-            rec["payload_off"] = 0
-            rec["flags"] = flags
-            if vis:
-                rec["vis_mode"] = VIS_BITMAP
-                rec["bitmap"] = self._bitmap_for(vis)
-            else:
-                rec["vis_mode"] = VIS_ALL
-                rec["bitmap"] = NO_BITMAP
+            rec, payload, overflow = self._encode_message(msg)
 
             try:
                 seqs = self.engine.enqueue_batch(rec, payload)
@@ -333,11 +299,114 @@ class SwarmsDB:
                 self._failed[msg.id] = msg
                 logger.error("send failed for %s: %s", msg.id, e)
                 raise
-            self._id_to_seq[msg.id] = int(seqs[0])
-            if overflow_payload is not None:
-                self._overflow[int(seqs[0])] = (len(content_b), overflow_payload)
+            self._register_enqueued(msg, int(seqs[0]), overflow)
             self._maybe_autosave()
             return msg.id
+
+    def _encode_message(self, msg: Message):
+        """Build the engine record for a validated Message. Returns
+        (rec[1], payload bytes, overflow tuple or None). Caller holds
+        the lock; sender/receiver must be registered."""
+        rec = np.zeros(1, dtype=REC_DTYPE)
+        content_b, is_json = encode_content(msg.content)
+        extras_b = encode_extras(msg.id, msg.metadata, msg.visible_to)
+        payload = content_b + extras_b
+        flags = FLAG_HAS_EXTRAS
+        if is_json:
+            flags |= FLAG_JSON_CONTENT
+        overflow = None
+        if len(payload) > self._slot_capacity():
+            # oversized content: the device slot stays empty and the
+            # payload lives host-side; routing/delivery is unchanged
+            overflow = (len(content_b), payload)
+            payload = b""
+            flags |= FLAG_OVERFLOW
+        else:
+            rec["payload_len"] = len(payload)
+            rec["content_len"] = len(content_b)
+        rec["sender"] = self._agent_idx[msg.sender_id]
+        rec["receiver"] = (
+            BROADCAST
+            if msg.receiver_id is None
+            else self._agent_idx[msg.receiver_id]
+        )
+        rec["type"] = _type_code(msg.type)
+        rec["priority"] = msg.priority.value
+        rec["timestamp"] = msg.timestamp
+        rec["token_count"] = msg.token_count or 0
+        rec["payload_off"] = 0
+        rec["flags"] = flags
+        if msg.visible_to:
+            rec["vis_mode"] = VIS_BITMAP
+            rec["bitmap"] = self._bitmap_for(msg.visible_to)
+        else:
+            rec["vis_mode"] = VIS_ALL
+            rec["bitmap"] = NO_BITMAP
+        return rec, payload, overflow
+
+    def _register_enqueued(self, msg: Message, seq: int, overflow) -> None:
+        self._id_to_seq[msg.id] = seq
+        if overflow is not None:
+            self._overflow[seq] = overflow
+
+    def send_messages_bulk(self, msgs: List[Message]) -> List[str]:
+        """Enqueue many validated Messages as ONE engine batch (the
+        micro-batcher's flush path — SURVEY.md §7 hard part 3: batched
+        submission across concurrent API requests). Senders/receivers
+        must be registered; returns message ids in order."""
+        with self._lock:
+            recs = []
+            payloads = []
+            overflows = []
+            off = 0
+            for m in msgs:
+                rec, payload, overflow = self._encode_message(m)
+                pad = (-len(payload)) % 16
+                rec["payload_off"] = off
+                recs.append(rec)
+                payloads.append(payload + b"\x00" * pad)
+                overflows.append(overflow)
+                off += len(payload) + pad
+            batch = np.concatenate(recs) if recs else np.empty(0, dtype=REC_DTYPE)
+            seqs = self.engine.enqueue_batch(batch, b"".join(payloads))
+            for m, s, ov in zip(msgs, seqs, overflows):
+                self._register_enqueued(m, int(s), ov)
+        self._maybe_autosave()
+        return [m.id for m in msgs]
+
+    def make_message(
+        self,
+        sender_id: str,
+        content: Union[str, Dict[str, Any], List[Any]],
+        receiver_id: Optional[str] = None,
+        message_type: Union[MessageType, str] = MessageType.CHAT,
+        priority: Union[MessagePriority, int] = MessagePriority.NORMAL,
+        metadata: Optional[Dict[str, Any]] = None,
+        visible_to: Optional[List[str]] = None,
+    ) -> Message:
+        """Validate + register, producing a Message ready for
+        send_messages_bulk (same semantics as send_message's prologue)."""
+        with self._lock:
+            self.register_agent(sender_id)
+            if receiver_id is not None:
+                self.register_agent(receiver_id)
+            vis = list(visible_to) if visible_to else []
+            if receiver_id is None and not vis:
+                vis = sorted(self.registered_agents)
+            return Message(
+                sender_id=sender_id,
+                receiver_id=receiver_id,
+                content=content,
+                type=MessageType(message_type),
+                priority=(
+                    priority
+                    if isinstance(priority, MessagePriority)
+                    else MessagePriority(priority)
+                ),
+                metadata=metadata or {},
+                token_count=self._count_tokens(content),
+                visible_to=vis,
+            )
 
     def broadcast_message(
         self,

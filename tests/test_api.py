@@ -392,3 +392,50 @@ def test_search_endpoint(client):
     assert len(r.json()) == 2
     # no auth
     assert client.get("/messages/search/?keyword=x").status_code == 401
+
+
+def test_micro_batched_sends(tmp_path):
+    """With a batch window, concurrent sends coalesce into one engine
+    batch; responses stay per-request correct."""
+    import threading
+
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path / "h"),
+                      max_agents=256, auto_save=False)
+    db = SwarmsDB(config=cfg)
+    calls = {"n": 0}
+    orig = db.engine.enqueue_batch
+
+    def counting(recs, payloads):
+        calls["n"] += 1
+        return orig(recs, payloads)
+
+    db.engine.enqueue_batch = counting
+    app = create_app(db=db, settings=ApiSettings(), batch_window=0.01)
+    with TestClient(app) as c:
+        tok = c.post("/auth/token",
+                     json={"username": "alice", "password": "x"}).json()
+        h = {"Authorization": f"Bearer {tok['access_token']}"}
+        results = []
+
+        def one(i):
+            r = c.post("/messages", headers=h,
+                       json={"receiver_id": "bob", "content": f"m{i}"})
+            results.append(r)
+
+        threads = [threading.Thread(target=one, args=(i,)) for i in range(12)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert all(r.status_code == 200 for r in results)
+        ids = {r.json()["id"] for r in results}
+        assert len(ids) == 12
+        # engine batches << request count (coalescing happened)
+        assert calls["n"] < 12, calls["n"]
+        tok2 = c.post("/auth/token",
+                      json={"username": "bob", "password": "x"}).json()
+        h2 = {"Authorization": f"Bearer {tok2['access_token']}"}
+        got = c.post("/agents/receive?timeout=0&max_messages=100",
+                     headers=h2).json()
+        assert {m["id"] for m in got} == ids
+    db.config.auto_save = False
