@@ -245,6 +245,80 @@ def config6_search(seconds: float) -> None:
     }))
 
 
+def config8_hbm_scale(seconds: float) -> None:
+    """Bonus: HBM-scale residency — fill a 64 GB / 32M-message log in
+    device memory (the 288 GB HBM3E is the store, not a cache) and run
+    full-content search + filtered queries against it."""
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    n_msgs = 1 << 25          # 32M messages
+    slot = 2048               # 64 GiB of payload slots
+    plen = 1984
+    cfg = QueueConfig(use_gpu=True, auto_save=False, max_agents=64,
+                      num_slots=n_msgs, slot_bytes=slot,
+                      staging_batch=1 << 16, inbox_capacity=1 << 10)
+    eng = GpuEngine(cfg)
+    rng = np.random.default_rng(0)
+    agents = np.arange(64, dtype=np.uint32)
+    for a in agents:
+        eng.register_agent(int(a))
+
+    batch = 1 << 16
+    recs, payload = _make_batch(rng, batch, agents, agents, plen)
+    pay = bytearray(payload)
+    stride = (plen + 15) // 16 * 16
+    for i in range(0, batch, 1024):   # needle in 1/1024 messages
+        off = i * stride + 500
+        pay[off : off + 10] = b"DEEPNEEDLE"
+    payload = bytes(pay)
+
+    pr = eng.q.alloc_pinned(recs.nbytes)
+    np.frombuffer(pr, dtype=np.uint8)[:] = np.frombuffer(
+        np.ascontiguousarray(recs).tobytes(), np.uint8)
+    pp = eng.q.alloc_pinned(len(payload))
+    np.frombuffer(pp, dtype=np.uint8)[:] = np.frombuffer(payload, np.uint8)
+
+    t0 = time.perf_counter()
+    nfill = n_msgs // batch
+    for it in range(nfill):
+        s = it % 2
+        eng.q.prefetch_from(s, pr.__array_interface__["data"][0],
+                            pp.__array_interface__["data"][0], batch,
+                            len(payload))
+        eng.q.enqueue_staged(s)
+    eng.q.sync()
+    fill_s = time.perf_counter() - t0
+    resident_gb = n_msgs * slot / 2**30
+
+    lat = []
+    nsearch = 0
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        hits = eng.search(b"DEEPNEEDLE", case_sensitive=True, limit=4096)
+        lat.append(time.perf_counter() - s)
+        assert len(hits) > 0
+        nsearch += 1
+    qlat = []
+    for _ in range(10):
+        s = time.perf_counter()
+        eng.query(sender=3, limit=10000)
+        qlat.append(time.perf_counter() - s)
+    eng.close()
+    print(json.dumps({
+        "config": 8,
+        "name": "hbm-scale-64gb-log",
+        "resident_messages": n_msgs,
+        "resident_slots_gb": round(resident_gb, 1),
+        "fill_s": round(fill_s, 2),
+        "fill_msgs_per_s": round(n_msgs / fill_s, 0),
+        "search_p50_ms": round(float(np.median(lat)) * 1000, 2),
+        "scan_tb_per_s": round(resident_gb / 1024 / float(np.median(lat)), 2),
+        "query_p50_ms": round(float(np.median(qlat)) * 1000, 2),
+        "searches": nsearch,
+    }))
+
+
 def config7_checkpoint(seconds: float) -> None:
     """Bonus: checkpoint/restore throughput — JSON history spill of a
     GPU-resident store (device->pinned gather on the copy stream + host
@@ -356,7 +430,7 @@ def config5_loadbalancer(seconds: float) -> None:
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--config", type=int, required=True,
-                    choices=[1, 2, 3, 4, 5, 6, 7])
+                    choices=[1, 2, 3, 4, 5, 6, 7, 8])
     ap.add_argument("--seconds", type=float, default=5.0)
     args = ap.parse_args()
     if args.config == 1:
@@ -383,6 +457,8 @@ def main() -> int:
         config6_search(args.seconds)
     elif args.config == 7:
         config7_checkpoint(args.seconds)
+    elif args.config == 8:
+        config8_hbm_scale(args.seconds)
     return 0
 
 
