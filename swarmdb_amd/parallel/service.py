@@ -375,6 +375,23 @@ class DistributedSwarmsDB(SwarmsDB):
         with self._lock:
             return self._pending_meta.get(message_id)
 
+    def send_to_group_fast(self, group_name, sender_id, content,
+                           message_type=MessageType.CHAT,
+                           priority=MessagePriority.NORMAL,
+                           metadata=None):  # type: ignore[override]
+        """The single-slot group fan-out is a per-shard engine
+        operation; in the distributed service group sends go through the
+        tick exchange as per-member messages instead (one id per member,
+        reference semantics)."""
+        ids = self.send_to_group(group_name, sender_id, content,
+                                 message_type=message_type,
+                                 priority=priority, metadata=metadata)
+        return ids[0] if ids else None
+
+    # NOTE: query/search/stats/history operate on THIS rank's shard.
+    # Aggregate views are the caller's concern (run them on every rank
+    # and merge); the data plane keeps each message on exactly one rank.
+
     def close(self) -> None:  # type: ignore[override]
         self.stop_ticker()
         super().close()
