@@ -89,8 +89,9 @@ def main() -> int:
                     help="skip payload D2H gather (delivery stays device-side)")
     ap.add_argument("--dump-steps", action="store_true",
                     help="print per-step wall times (variance diagnosis)")
-    ap.add_argument("--no-graph", action="store_true",
-                    help="disable the hipGraph-captured tick")
+    ap.add_argument("--graph", action="store_true",
+                    help="use the hipGraph-captured tick (measured neutral "
+                         "at large batches, slower at small — default off)")
     args = ap.parse_args()
 
     import torch
@@ -257,7 +258,7 @@ def main() -> int:
 
         # hipGraph-captured steady-state tick: enqueue + fanout +
         # receive + result D2H replay as ONE graph launch per step
-        use_graph = not args.no_graph
+        use_graph = args.graph
         if use_graph:
             try:
                 q.build_tick(args.batch, local_agents.astype(np.uint32),

@@ -290,6 +290,15 @@ def config8_hbm_scale(seconds: float) -> None:
     fill_s = time.perf_counter() - t0
     resident_gb = n_msgs * slot / 2**30
 
+    # (a) FULL scan: absent needle forces a pass over every content byte
+    full_lat = []
+    for _ in range(5):
+        s = time.perf_counter()
+        hits = eng.search(b"NO-SUCH-NEEDLE-ANYWHERE", case_sensitive=True,
+                          limit=4096)
+        full_lat.append(time.perf_counter() - s)
+        assert len(hits) == 0
+    # (b) early-exit: stop once `limit` planted needles are found
     lat = []
     nsearch = 0
     t0 = time.perf_counter()
@@ -305,6 +314,7 @@ def config8_hbm_scale(seconds: float) -> None:
         eng.query(sender=3, limit=10000)
         qlat.append(time.perf_counter() - s)
     eng.close()
+    full_p50 = float(np.median(full_lat))
     print(json.dumps({
         "config": 8,
         "name": "hbm-scale-64gb-log",
@@ -312,8 +322,9 @@ def config8_hbm_scale(seconds: float) -> None:
         "resident_slots_gb": round(resident_gb, 1),
         "fill_s": round(fill_s, 2),
         "fill_msgs_per_s": round(n_msgs / fill_s, 0),
-        "search_p50_ms": round(float(np.median(lat)) * 1000, 2),
-        "scan_tb_per_s": round(resident_gb / 1024 / float(np.median(lat)), 2),
+        "full_scan_p50_ms": round(full_p50 * 1000, 2),
+        "full_scan_tb_per_s": round(resident_gb / 1024 / full_p50, 2),
+        "first4096_hits_p50_ms": round(float(np.median(lat)) * 1000, 2),
         "query_p50_ms": round(float(np.median(qlat)) * 1000, 2),
         "searches": nsearch,
     }))
