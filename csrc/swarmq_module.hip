@@ -455,10 +455,16 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
   const u32 cps = g.slot_bytes >> 4; // chunks per slot
   const u64 nchunks = (hi - lo) * cps;
   const u64 stride = (u64)gridDim.x * blockDim.x;
-  for (u64 ci = (u64)blockIdx.x * blockDim.x + threadIdx.x; ci < nchunks;
-       ci += stride) {
-    const u64 seq = lo + ci / cps;
-    const u32 sub = (u32)(ci % cps);
+  const u64 ci0 = (u64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (ci0 >= nchunks)
+    return;
+  // incremental (seq, sub) stepping: one division at entry, none in the
+  // loop (a per-chunk 64-bit div/mod would dominate the scan's VALU)
+  u64 seq = lo + ci0 / cps;
+  u32 sub = (u32)(ci0 % cps);
+  const u64 dseq = stride / cps;
+  const u32 dsub = (u32)(stride % cps);
+  for (u64 ci = ci0; ci < nchunks; ci += stride) {
     const u32 slot = (u32)(seq % g.num_slots);
     const uint4 v = reinterpret_cast<const uint4 *>(
         payload + (u64)slot * g.slot_bytes)[sub];
@@ -467,38 +473,40 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
 #pragma unroll
     for (int j = 0; j < 16; ++j)
       hitmask |= (fold_c(b[j], fold) == n0) ? (1u << j) : 0u;
-    if (hitmask == 0)
-      continue;
-    // candidate path (rare): check the message and verify positions
-    const Rec h = hdr[slot];
-    if (status[slot] == ST_DELETED)
-      continue;
-    const int clen = (int)h.content_len;
-    const int nstart = clen - nlen + 1;
-    if (nstart <= 0)
-      continue;
-    const u8 *text = payload + (u64)slot * g.slot_bytes;
-    const int base = (int)(sub << 4);
     bool found = false;
-    while (hitmask && !found) {
-      const int j = __builtin_ctz(hitmask);
-      hitmask &= hitmask - 1;
-      const int p = base + j;
-      if (p >= nstart)
-        continue;
-      bool m = true;
-      for (int q = 1; q < nlen; ++q) {
-        if (fold_c(text[p + q], fold) != fold_c(needle[q], fold)) {
-          m = false;
-          break;
+    if (hitmask != 0 && status[slot] != ST_DELETED) {
+      // candidate path (rare): check the message and verify positions;
+      // the trailing bytes come from L1/L2 (the wave just streamed them)
+      const Rec h = hdr[slot];
+      const int nstart = (int)h.content_len - nlen + 1;
+      const u8 *text = payload + (u64)slot * g.slot_bytes;
+      const int base = (int)(sub << 4);
+      while (hitmask && !found && nstart > 0) {
+        const int j = __builtin_ctz(hitmask);
+        hitmask &= hitmask - 1;
+        const int p = base + j;
+        if (p >= nstart)
+          continue;
+        bool m = true;
+        for (int q = 1; q < nlen; ++q) {
+          if (fold_c(text[p + q], fold) != fold_c(needle[q], fold)) {
+            m = false;
+            break;
+          }
         }
+        found = m;
       }
-      found = m;
+      if (found) {
+        const u32 i = atomicAdd(out_count, 1u);
+        if (i < cap)
+          out[i] = seq;
+      }
     }
-    if (found) {
-      const u32 i = atomicAdd(out_count, 1u);
-      if (i < cap)
-        out[i] = seq;
+    seq += dseq;
+    sub += dsub;
+    if (sub >= cps) {
+      sub -= cps;
+      ++seq;
     }
   }
 }
