@@ -245,6 +245,71 @@ def config6_search(seconds: float) -> None:
     }))
 
 
+def config9_distributed_service(seconds: float) -> None:
+    """Distributed service tick throughput (DistributedSwarmsDB).
+    Launch under torch.distributed.run; gloo on CPU, RCCL on GPUs:
+
+      python -m torch.distributed.run --nproc-per-node 2 \
+        --master-addr 127.0.0.1 benchmarks/run.py --config 9
+    """
+    import torch
+    import torch.distributed as dist
+
+    from swarmdb_amd.parallel.service import DistributedSwarmsDB
+
+    if not dist.is_initialized():
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    cfg = QueueConfig(auto_save=False, max_agents=1024,
+                      use_gpu=torch.cuda.is_available())
+    svc = DistributedSwarmsDB(config=cfg)
+    agents = [f"agent{i}" for i in range(256)]
+    for a in agents:
+        svc.register_agent(a)
+    svc.tick()
+    local = [a for a in agents if svc.is_local(a)]
+    rng = np.random.default_rng(rank)
+
+    sent = 0
+    delivered = 0
+    ticks = 0
+    lat = []
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < seconds:
+        s = time.perf_counter()
+        for _ in range(64):  # per-message API sends per tick per rank
+            snd = local[int(rng.integers(0, len(local)))]
+            rcv = agents[int(rng.integers(0, len(agents)))]
+            if rcv != snd:
+                svc.send_message(snd, "tick payload " + "x" * 200,
+                                 receiver_id=rcv)
+                sent += 1
+        svc.tick()
+        for a in local:
+            delivered += len(svc.receive_messages(a, timeout=0))
+        lat.append(time.perf_counter() - s)
+        ticks += 1
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([float(sent), float(delivered)])
+    dist.all_reduce(t)
+    if rank == 0:
+        print(json.dumps({
+            "config": 9,
+            "name": "distributed-service",
+            "world": world,
+            "backend": dist.get_backend(),
+            "messages_per_s": round(float(t[1]) / elapsed, 1),
+            "tick_p50_ms": round(float(np.median(lat)) * 1000, 2),
+            "ticks": ticks,
+            "sent": int(t[0]),
+            "delivered": int(t[1]),
+        }))
+    svc.config.auto_save = False
+    dist.destroy_process_group()
+
+
 def config8_hbm_scale(seconds: float) -> None:
     """Bonus: HBM-scale residency — fill a 64 GB / 32M-message log in
     device memory (the 288 GB HBM3E is the store, not a cache) and run
@@ -441,7 +506,7 @@ def config5_loadbalancer(seconds: float) -> None:
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--config", type=int, required=True,
-                    choices=[1, 2, 3, 4, 5, 6, 7, 8])
+                    choices=[1, 2, 3, 4, 5, 6, 7, 8, 9])
     ap.add_argument("--seconds", type=float, default=5.0)
     args = ap.parse_args()
     if args.config == 1:
@@ -470,6 +535,8 @@ def main() -> int:
         config7_checkpoint(args.seconds)
     elif args.config == 8:
         config8_hbm_scale(args.seconds)
+    elif args.config == 9:
+        config9_distributed_service(args.seconds)
     return 0
 
 
