@@ -506,23 +506,11 @@ class SwarmsDB:
                 token_count=self._count_tokens(content),
                 visible_to=members,
             )
-            rec = np.zeros(1, dtype=REC_DTYPE)
-            content_b, is_json = encode_content(content)
-            extras_b = encode_extras(msg.id, md, members)
-            payload = content_b + extras_b
-            rec["sender"] = self._agent_idx[sender_id]
-            rec["receiver"] = BROADCAST
-            rec["type"] = _type_code(mtype)
-            rec["priority"] = prio.value
-            rec["timestamp"] = msg.timestamp
-            rec["token_count"] = msg.token_count or 0
-            rec["payload_len"] = len(payload)
-            rec["content_len"] = len(content_b)
-            rec["flags"] = FLAG_HAS_EXTRAS | (FLAG_JSON_CONTENT if is_json else 0)
+            rec, payload, overflow = self._encode_message(msg)
             rec["vis_mode"] = VIS_GROUP
             rec["bitmap"] = self._group_bitmap(group_name, members)
             seqs = self.engine.enqueue_batch(rec, payload)
-            self._id_to_seq[msg.id] = int(seqs[0])
+            self._register_enqueued(msg, int(seqs[0]), overflow)
             self._maybe_autosave()
             return msg.id
 
