@@ -303,10 +303,11 @@ class SwarmsDB:
             self._maybe_autosave()
             return msg.id
 
-    def _encode_message(self, msg: Message):
+    def _encode_message(self, msg: Message, bitmap_idx: Optional[int] = None):
         """Build the engine record for a validated Message. Returns
         (rec[1], payload bytes, overflow tuple or None). Caller holds
-        the lock; sender/receiver must be registered."""
+        the lock; sender/receiver must be registered. ``bitmap_idx``
+        reuses a pre-allocated visibility bitmap (group cache)."""
         rec = np.zeros(1, dtype=REC_DTYPE)
         content_b, is_json = encode_content(msg.content)
         extras_b = encode_extras(msg.id, msg.metadata, msg.visible_to)
@@ -338,7 +339,10 @@ class SwarmsDB:
         rec["flags"] = flags
         if msg.visible_to:
             rec["vis_mode"] = VIS_BITMAP
-            rec["bitmap"] = self._bitmap_for(msg.visible_to)
+            rec["bitmap"] = (
+                bitmap_idx if bitmap_idx is not None
+                else self._bitmap_for(msg.visible_to)
+            )
         else:
             rec["vis_mode"] = VIS_ALL
             rec["bitmap"] = NO_BITMAP
@@ -506,9 +510,10 @@ class SwarmsDB:
                 token_count=self._count_tokens(content),
                 visible_to=members,
             )
-            rec, payload, overflow = self._encode_message(msg)
+            rec, payload, overflow = self._encode_message(
+                msg, bitmap_idx=self._group_bitmap(group_name, members)
+            )
             rec["vis_mode"] = VIS_GROUP
-            rec["bitmap"] = self._group_bitmap(group_name, members)
             seqs = self.engine.enqueue_batch(rec, payload)
             self._register_enqueued(msg, int(seqs[0]), overflow)
             self._maybe_autosave()

@@ -56,6 +56,10 @@ class GpuEngine(Engine):
             slot_bytes=c.slot_bytes,
             max_agents=c.max_agents,
             inbox_capacity=c.inbox_capacity,
+            # visibility bitmaps live in a ring pool: the most recent
+            # 4096 restricted broadcasts keep exact visibility; older
+            # unread restricted broadcasts may see a recycled bitmap
+            # (retention analog of the slot ring)
             num_bitmaps=4096,
             num_backends=c.num_backends,
             staging_batch=c.staging_batch,
