@@ -464,6 +464,7 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
   u32 sub = (u32)(ci0 % cps);
   const u64 dseq = stride / cps;
   const u32 dsub = (u32)(stride % cps);
+#pragma unroll 2
   for (u64 ci = ci0; ci < nchunks; ci += stride) {
     const u32 slot = (u32)(seq % g.num_slots);
     const uint4 v = reinterpret_cast<const uint4 *>(
@@ -1414,7 +1415,7 @@ public:
       HIP_CHECK(hipMemsetAsync(d_match_count_, 0, sizeof(u32), stream_));
       const u64 span = hi - lo;
       const u64 chunks = span * (g_.slot_bytes >> 4);
-      const int blocks = (int)std::min<u64>((chunks + 255) / 256, 2048);
+      const int blocks = (int)std::min<u64>((chunks + 255) / 256, 4096);
       hipLaunchKernelGGL(k_search, dim3(blocks), dim3(256), 0, stream_, lo, hi,
                          d_needle_, (int)nd.size(), fold ? 1 : 0, d_hdr_,
                          d_status_, d_payload_, d_match_, d_match_count_, cap,
