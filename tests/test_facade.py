@@ -430,3 +430,30 @@ def test_oversized_content_overflow_store(tmp_db):
     mid2 = tmp_db.send_message("a", big_dict, receiver_id="b")
     assert tmp_db.receive_messages("b", timeout=0)[0].content == big_dict
     assert tmp_db.get_message(mid2).content == big_dict
+
+
+def test_receive_timeout_waits_for_message(tmp_db):
+    """receive_messages(timeout>0) polls until a message arrives
+    (reference consumer-poll semantics, swarmdb/ main.py:553-558)."""
+    import threading
+    import time as _t
+
+    tmp_db.register_agent("waiter")
+
+    def late_send():
+        _t.sleep(0.15)
+        tmp_db.send_message("someone", "late delivery", receiver_id="waiter")
+
+    t = threading.Thread(target=late_send)
+    start = _t.monotonic()
+    t.start()
+    msgs = tmp_db.receive_messages("waiter", timeout=2.0)
+    elapsed = _t.monotonic() - start
+    t.join()
+    assert len(msgs) == 1
+    assert msgs[0].content == "late delivery"
+    assert 0.1 < elapsed < 1.5  # returned as soon as it arrived
+    # and an empty timeout expires without messages
+    start = _t.monotonic()
+    assert tmp_db.receive_messages("waiter", timeout=0.1) == []
+    assert _t.monotonic() - start >= 0.09
