@@ -310,7 +310,9 @@ class GpuEngine(Engine):
         nfound = 0
         hi = total
         chunk = max(1 << 20, limit)
-        # scan newest chunks first until the limit fills
+        # scan newest chunks first until the limit fills; if a window
+        # overflows the match buffer the subset is arbitrary, so shrink
+        # the window until it fits (newest-first stays exact)
         while hi > eb and nfound < limit:
             lo = max(eb, hi - chunk)
             seqs = self.q.query_range(
@@ -325,6 +327,9 @@ class GpuEngine(Engine):
                 mask,
                 self._staging,
             )
+            if len(seqs) >= self._staging and hi - lo > 1:
+                chunk = max(1, chunk // 4)
+                continue
             seqs = np.sort(np.asarray(seqs, dtype=np.uint64))[::-1]
             found.append(seqs)
             nfound += len(seqs)
@@ -346,6 +351,9 @@ class GpuEngine(Engine):
             seqs = self.q.search_range(
                 lo, hi, needle, not case_sensitive, self._staging
             )
+            if len(seqs) >= self._staging and hi - lo > 1:
+                chunk = max(1, chunk // 4)
+                continue
             # the linear-scan kernel may report a message once per
             # matching chunk: dedup
             seqs = np.unique(np.asarray(seqs, dtype=np.uint64))[::-1]
