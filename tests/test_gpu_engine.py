@@ -667,3 +667,21 @@ def test_graph_tick_matches_discrete_path(gpu_engine):
     # graph path kept counters coherent
     stats = eng.stats_arrays()
     assert stats["by_status"][ST_READ] == total
+
+
+def test_query_overflowing_window_stays_newest_first(gpu_engine):
+    """When matches exceed the device match buffer, the window shrinks
+    and newest-first semantics stay exact."""
+    eng = gpu_engine
+    eng.register_agent(0)
+    eng.register_agent(1)
+    rng = np.random.default_rng(17)
+    total = 10000  # >> staging_batch (4096) matching messages
+    for _ in range(total // 2000):
+        recs, payload = make_batch(rng, 2000, 1, payload_bytes=32)
+        recs["sender"] = 0
+        recs["receiver"] = 1
+        eng.enqueue_batch(recs, payload)
+    got = eng.query(sender=0, limit=3000)
+    expect = np.arange(total - 1, total - 3001, -1, dtype=np.uint64)
+    assert (got == expect).all()
