@@ -850,16 +850,7 @@ public:
                                  hipMemcpyHostToDevice, stream_));
       HIP_CHECK(hipEventRecord(stage_ev_[0], stream_));
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
-                         stream_, d_stage_recs_[0], n, base, (const u64 *)nullptr,
-                         d_hdr_, d_status_, d_inbox_, d_wpos_, d_by_type_,
-                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
-      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
-                         stream_, d_stage_recs_[0], d_stage_pay_[0], n, base,
-                         (const u64 *)nullptr, d_payload_, g_);
-      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
-                         0, stream_, d_bcast_, d_bcast_count_, d_active_,
-                         d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
+      launch_enqueue(d_stage_recs_[0], d_stage_pay_[0], n, base);
       HIP_CHECK(hipStreamSynchronize(stream_));
     }
     count_ = base + (u64)n;
@@ -994,18 +985,7 @@ public:
         HIP_CHECK(hipEventRecord(stage_ev_[slot], stream_));
       }
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
-                         stream_, d_stage_recs_[slot], n, base,
-                         (const u64 *)nullptr, d_hdr_, d_status_, d_inbox_,
-                         d_wpos_, d_by_type_, d_by_status_, d_sent_, d_bcast_,
-                         d_bcast_count_, g_);
-      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
-                         stream_, d_stage_recs_[slot], d_stage_pay_[slot], n,
-                         base, (const u64 *)nullptr, d_payload_, g_);
-      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
-                         dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
-                         d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
-                         g_);
+      launch_enqueue(d_stage_recs_[slot], d_stage_pay_[slot], n, base);
     }
     count_ = base + (u64)n;
     if (count_ > g_.num_slots)
@@ -1145,19 +1125,8 @@ public:
     {
       py::gil_scoped_release nogil;
       HIP_CHECK(hipMemsetAsync(d_bcast_count_, 0, sizeof(u32), stream_));
-      hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
-                         stream_, reinterpret_cast<const Rec *>(recs_ptr), n,
-                         base, (const u64 *)nullptr, d_hdr_, d_status_,
-                         d_inbox_, d_wpos_, d_by_type_, d_by_status_, d_sent_,
-                         d_bcast_, d_bcast_count_, g_);
-      hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
-                         stream_, reinterpret_cast<const Rec *>(recs_ptr),
-                         reinterpret_cast<const u8 *>(pay_ptr), n, base,
-                         (const u64 *)nullptr, d_payload_, g_);
-      hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
-                         dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
-                         d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
-                         g_);
+      launch_enqueue(reinterpret_cast<const Rec *>(recs_ptr),
+                     reinterpret_cast<const u8 *>(pay_ptr), n, base);
     }
     count_ = base + (u64)n;
     if (count_ > g_.num_slots)
@@ -1542,6 +1511,22 @@ public:
   u32 recv_window() const { return g_.recv_window; }
 
 private:
+  // Launch the enqueue kernel pair (+ fan-out) for n records at
+  // `recs`/`pay` (device pointers) with host-known base seq; async on
+  // stream_. Callers already reset d_bcast_count_.
+  void launch_enqueue(const Rec *recs, const u8 *pay, int n, u64 base) {
+    hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
+                       stream_, recs, n, base, (const u64 *)nullptr, d_hdr_,
+                       d_status_, d_inbox_, d_wpos_, d_by_type_, d_by_status_,
+                       d_sent_, d_bcast_, d_bcast_count_, g_);
+    hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
+                       stream_, recs, pay, n, base, (const u64 *)nullptr,
+                       d_payload_, g_);
+    hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
+                       0, stream_, d_bcast_, d_bcast_count_, d_active_,
+                       d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
+  }
+
   void check_agent(u32 idx) const {
     if (idx >= g_.max_agents)
       throw std::out_of_range("agent index out of range");
