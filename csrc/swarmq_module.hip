@@ -354,6 +354,50 @@ __global__ void __launch_bounds__(256)
   }
 }
 
+// Prefix-scan the per-agent delivery counts into dense output offsets
+// (single workgroup; agents <= max_agents).
+__global__ void k_scan_offsets(const u32 *__restrict__ counts, int n,
+                               u32 *__restrict__ offsets,
+                               u32 *__restrict__ total) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    u32 acc = 0;
+    for (int i = 0; i < n; ++i) {
+      offsets[i] = acc;
+      acc += counts[i];
+    }
+    *total = acc;
+  }
+}
+
+// Gather delivered payloads straight from the dequeue output buffer
+// (device-resident) into a dense D2H staging area — the delivery path
+// never round-trips seqs through the host.
+__global__ void k_gather_outbuf(const u64 *__restrict__ out_seqs,
+                                const u32 *__restrict__ counts,
+                                const u32 *__restrict__ offsets, int n_agents,
+                                int K, const Rec *__restrict__ hdr,
+                                const u8 *__restrict__ payload,
+                                u8 *__restrict__ out_pay, u32 stride,
+                                QueueGeom g) {
+  const int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  const int a = wave / K;
+  const int i = wave % K;
+  if (a >= n_agents || (u32)i >= counts[a])
+    return;
+  const u64 seq = out_seqs[(u64)a * K + i];
+  const Rec h = hdr[seq % g.num_slots];
+  const uint4 *src = reinterpret_cast<const uint4 *>(payload + h.payload_off);
+  uint4 *dst = reinterpret_cast<uint4 *>(
+      out_pay + (u64)(offsets[a] + i) * stride);
+  u32 plen = h.payload_len;
+  if (plen > stride)
+    plen = stride;
+  const u32 nchunk = (plen + 15u) >> 4;
+  for (u32 c = lane; c < nchunk; c += 64)
+    dst[c] = src[c];
+}
+
 // Gather message contents for the host (receive payload delivery, fetch,
 // history spill): one wave per message, dense slot_bytes-strided output.
 __global__ void k_gather(const u64 *__restrict__ seqs, int n,
@@ -1070,6 +1114,8 @@ public:
     count_ += (u64)tick_n_;
     if (count_ > g_.num_slots)
       evict_base_ = count_ - g_.num_slots;
+    last_recv_na_ = tick_na_;
+    last_recv_K_ = tick_K_;
     std::memcpy(counts.mutable_data(), h_out_counts_,
                 tick_na_ * sizeof(u32));
     std::memcpy(seqs.mutable_data(), h_out_seqs_,
@@ -1149,14 +1195,14 @@ public:
   // Returns (counts[n_agents], seqs[n_agents * max_per_agent]) — one
   // dequeue-kernel launch for the whole poll tick.
   py::tuple receive_many(py::array_t<u32> agents, int max_per_agent,
-                         bool priority) {
+                         bool priority, bool return_seqs = true) {
     const int na = (int)agents.size();
     if (na == 0)
       return py::make_tuple(py::array_t<u32>(0), py::array_t<u64>(0));
     if ((size_t)na * max_per_agent > out_pool_)
       throw std::invalid_argument("receive batch exceeds output pool");
     py::array_t<u32> counts(na);
-    py::array_t<u64> seqs((size_t)na * max_per_agent);
+    py::array_t<u64> seqs(return_seqs ? (size_t)na * max_per_agent : 0);
     {
       py::gil_scoped_release nogil;
       HIP_CHECK(hipMemcpyAsync(d_agents_, agents.data(), na * sizeof(u32),
@@ -1169,15 +1215,56 @@ public:
                          d_out_counts_, d_by_status_, d_received_, g_);
       HIP_CHECK(hipMemcpyAsync(h_out_counts_, d_out_counts_, na * sizeof(u32),
                                hipMemcpyDeviceToHost, stream_));
-      HIP_CHECK(hipMemcpyAsync(h_out_seqs_, d_out_seqs_,
-                               (size_t)na * max_per_agent * sizeof(u64),
-                               hipMemcpyDeviceToHost, stream_));
+      if (return_seqs)
+        HIP_CHECK(hipMemcpyAsync(h_out_seqs_, d_out_seqs_,
+                                 (size_t)na * max_per_agent * sizeof(u64),
+                                 hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipStreamSynchronize(stream_));
     }
     std::memcpy(counts.mutable_data(), h_out_counts_, na * sizeof(u32));
-    std::memcpy(seqs.mutable_data(), h_out_seqs_,
-                (size_t)na * max_per_agent * sizeof(u64));
+    if (return_seqs)
+      std::memcpy(seqs.mutable_data(), h_out_seqs_,
+                  (size_t)na * max_per_agent * sizeof(u64));
+    last_recv_na_ = na;
+    last_recv_K_ = max_per_agent;
     return py::make_tuple(counts, seqs);
+  }
+
+  // Deliver the payloads of the LAST receive_many/run_tick straight from
+  // the device-resident output buffer: scan counts -> dense gather ->
+  // one D2H of exactly `total` x stride bytes into pinned memory.
+  u64 deliver_outbuf(u32 total, u32 stride, bool synchronize) {
+    if (total == 0 || last_recv_na_ == 0)
+      return 0;
+    if (stride == 0 || stride > g_.slot_bytes)
+      stride = g_.slot_bytes;
+    stride = (stride + 15u) & ~15u;
+    if ((size_t)total * stride > (size_t)staging_batch_ * g_.slot_bytes)
+      throw std::invalid_argument("delivery exceeds the pinned buffer");
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipEventRecord(ev_, stream_));
+      HIP_CHECK(hipStreamWaitEvent(copy_stream_, ev_, 0));
+      hipLaunchKernelGGL(k_scan_offsets, dim3(1), dim3(64), 0, copy_stream_,
+                         d_out_counts_, last_recv_na_, d_unread_,
+                         d_match_count_);
+      const int waves = last_recv_na_ * last_recv_K_;
+      hipLaunchKernelGGL(k_gather_outbuf, dim3((waves + 3) / 4), dim3(256), 0,
+                         copy_stream_, d_out_seqs_, d_out_counts_, d_unread_,
+                         last_recv_na_, last_recv_K_, d_hdr_, d_payload_,
+                         d_fetch_pay_, stride, g_);
+      HIP_CHECK(hipMemcpyAsync(h_fetch_pay_, d_fetch_pay_,
+                               (size_t)total * stride, hipMemcpyDeviceToHost,
+                               copy_stream_));
+      if (synchronize) {
+        HIP_CHECK(hipStreamSynchronize(copy_stream_));
+      } else {
+        HIP_CHECK(hipEventRecord(d2h_ev_[d2h_cur_], copy_stream_));
+        d2h_cur_ ^= 1;
+        HIP_CHECK(hipEventSynchronize(d2h_ev_[d2h_cur_]));
+      }
+    }
+    return (u64)total * stride;
   }
 
   // ---- message store ----
@@ -1556,6 +1643,8 @@ private:
   u32 staging_batch_;
   u64 count_ = 0;
   u64 evict_base_ = 0;
+  int last_recv_na_ = 0;
+  int last_recv_K_ = 0;
   u32 bitmap_next_ = 0;
   bool released_ = false;
   size_t out_pool_ = 0;
@@ -1655,7 +1744,10 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
       .def("pack_exchange", &DeviceQueue::pack_exchange)
       .def("enqueue_from_ptrs", &DeviceQueue::enqueue_from_ptrs)
-      .def("receive_many", &DeviceQueue::receive_many)
+      .def("receive_many", &DeviceQueue::receive_many, py::arg("agents"),
+           py::arg("max_per_agent"), py::arg("priority"),
+           py::arg("return_seqs") = true)
+      .def("deliver_outbuf", &DeviceQueue::deliver_outbuf)
       .def("fetch", &DeviceQueue::fetch)
       .def("fetch_raw", &DeviceQueue::fetch_raw, py::arg("seqs"),
            py::arg("stride") = 0, py::arg("synchronize") = true)
