@@ -314,12 +314,18 @@ def create_app(
         messages: List[MessageRequest], current: str = Depends(get_current_agent)
     ):
         """Bulk ingestion (new — no reference analog): the whole list goes
-        through ONE engine enqueue (one pinned H2D + one kernel on GPU)."""
+        through ONE engine enqueue (one pinned H2D + one kernel on GPU);
+        ids are derived from (rank, seq) with zero per-message host
+        state."""
+        import time as _time
+
         import numpy as np
 
         from ..core.wire import derived_id, encode_content
         from ..runtime.engine import (
             BROADCAST,
+            FLAG_DERIVED_ID,
+            FLAG_JSON_CONTENT,
             NO_BITMAP,
             REC_DTYPE,
             VIS_ALL,
@@ -332,9 +338,8 @@ def create_app(
         recs = np.zeros(n, dtype=REC_DTYPE)
         chunks: List[bytes] = []
         off = 0
-        import time as _time
-
         now = _time.time()
+        type_codes = {t: i for i, t in enumerate(MessageType)}
         for i, req in enumerate(messages):
             content_b, is_json = encode_content(req.content)
             pad = (-len(content_b)) % 16
@@ -345,7 +350,7 @@ def create_app(
                 if req.receiver_id is None
                 else db.agent_index(req.receiver_id)
             )
-            recs["type"][i] = list(MessageType).index(req.message_type)
+            recs["type"][i] = type_codes[req.message_type]
             recs["priority"][i] = req.priority.value
             recs["timestamp"][i] = now
             recs["vis_mode"][i] = VIS_ALL
@@ -353,8 +358,6 @@ def create_app(
             recs["payload_off"][i] = off
             recs["payload_len"][i] = len(content_b)
             recs["content_len"][i] = len(content_b)
-            from ..runtime.engine import FLAG_DERIVED_ID, FLAG_JSON_CONTENT
-
             recs["flags"][i] = FLAG_DERIVED_ID | (
                 FLAG_JSON_CONTENT if is_json else 0
             )
