@@ -302,6 +302,10 @@ def main() -> int:
     for i in range(args.warmup):
         step(i)
 
+    # warmup deliveries must not count toward the timed total
+    sent_total = 0
+    recv_total = 0
+
     barrier_sync()
     t0 = time.perf_counter()
     step_times = []
