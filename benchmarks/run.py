@@ -113,7 +113,9 @@ def config2_p2p_gpu(seconds: float) -> None:
     eng.register_agent(0)
     eng.register_agent(1)
     rng = np.random.default_rng(0)
-    batch = 16384
+    # a single inbox drains at most one dequeue window (4096) per call,
+    # so the tick size matches it — larger batches would silently back up
+    batch = 4096
     recs, payload = _make_batch(rng, batch, np.array([0]), np.array([1]), 1024)
     agents = np.array([1], dtype=np.uint32)
     lat = []
@@ -129,6 +131,7 @@ def config2_p2p_gpu(seconds: float) -> None:
         counts, seqs = eng.receive_many(agents, batch)
         eng.deliver_payloads(seqs, 1024)
         lat.append(time.perf_counter() - s)
+        assert int(counts.sum()) == batch, "p2p tick failed to drain"
         n += int(counts.sum())
     elapsed = time.perf_counter() - t0
     eng.close()
