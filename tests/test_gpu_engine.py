@@ -116,25 +116,38 @@ def test_parity_cpu_vs_gpu():
         cpu.register_agent(a)
 
     try:
-        for round_i in range(6):
+        from swarmdb_amd.runtime.engine import VIS_GROUP
+
+        for round_i in range(14):
             n = int(rng.integers(50, 400))
             recs, payload = make_batch(
                 rng, n, n_agents, payload_bytes=128, bcast_frac=0.1
             )
-            # some restricted-visibility broadcasts (bitmap allocators run
-            # in lockstep on both engines, so indices agree)
+            # restricted-visibility broadcasts and group fan-outs (bitmap
+            # allocators run in lockstep on both engines, so indices agree)
             bcast_idx = np.flatnonzero(recs["receiver"] == BROADCAST)
-            for j in bcast_idx[: len(bcast_idx) // 2]:
+            for pos, j in enumerate(bcast_idx[: len(bcast_idx) * 2 // 3]):
                 bits = rng.random(cfg.max_agents) < 0.5
                 bits[n_agents:] = False
                 bg = gpu.alloc_bitmap(bits)
                 bc = cpu.alloc_bitmap(bits)
                 assert bg == bc
-                recs["vis_mode"][j] = VIS_BITMAP
+                recs["vis_mode"][j] = VIS_BITMAP if pos % 2 else VIS_GROUP
                 recs["bitmap"][j] = bg
+            # a few malformed records exercise the error lane in-stream
+            if round_i % 4 == 3 and n > 10:
+                recs["type"][3] = 99
+                recs["receiver"][7] = 4_000_000
             sg = gpu.enqueue_batch(recs, payload)
             sc = cpu.enqueue_batch(recs, payload)
             assert (sg == sc).all()
+
+            # mid-stream deletes (tombstones must filter identically)
+            if round_i % 3 == 2:
+                for _ in range(3):
+                    s = int(rng.integers(0, gpu.total_messages()))
+                    gpu.delete(s)
+                    cpu.delete(s)
 
             # interleaved receives: random subset of agents, random K
             polls = rng.permutation(n_agents)[: int(rng.integers(8, n_agents))]
