@@ -720,14 +720,11 @@ public:
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&h2d_stream_, hipStreamNonBlocking));
-    HIP_CHECK(hipStreamCreateWithFlags(&h2d_stream2_, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&ev_, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&up_ev_[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&up_ev_[1], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&d2h_ev_[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&d2h_ev_[1], hipEventDisableTiming));
-    HIP_CHECK(hipEventCreateWithFlags(&up2_ev_[0], hipEventDisableTiming));
-    HIP_CHECK(hipEventCreateWithFlags(&up2_ev_[1], hipEventDisableTiming));
 
     // device state
     HIP_CHECK(hipMalloc(&d_hdr_, (size_t)num_slots * sizeof(Rec)));
@@ -840,12 +837,9 @@ public:
     (void)hipEventDestroy(up_ev_[1]);
     (void)hipEventDestroy(d2h_ev_[0]);
     (void)hipEventDestroy(d2h_ev_[1]);
-    (void)hipEventDestroy(up2_ev_[0]);
-    (void)hipEventDestroy(up2_ev_[1]);
     (void)hipStreamDestroy(stream_);
     (void)hipStreamDestroy(copy_stream_);
     (void)hipStreamDestroy(h2d_stream_);
-    (void)hipStreamDestroy(h2d_stream2_);
   }
 
   // ---- registry ----
@@ -938,7 +932,6 @@ public:
       py::gil_scoped_release nogil;
       // don't overwrite pinned memory a previous H2D still reads
       HIP_CHECK(hipEventSynchronize(stage_ev_[slot]));
-      HIP_CHECK(hipEventSynchronize(up2_ev_[slot]));
       std::memcpy(h_recs_[slot], ri.ptr, (size_t)n * sizeof(Rec));
       if (pay_bytes)
         par_memcpy(h_pay_[slot], pi.ptr, pay_bytes);
@@ -971,25 +964,16 @@ public:
     ensure_stage_pay(pay_bytes + 16);
     {
       py::gil_scoped_release nogil;
-      // split the payload across two H2D streams: two SDMA engines
-      // move it concurrently
-      const size_t half = (pay_bytes / 2) & ~(size_t)255;
       HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot],
                                reinterpret_cast<void *>(recs_ptr),
                                (size_t)n * sizeof(Rec),
                                hipMemcpyHostToDevice, h2d_stream_));
-      if (half)
+      if (pay_bytes)
         HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot],
-                                 reinterpret_cast<void *>(pay_ptr), half,
-                                 hipMemcpyHostToDevice, h2d_stream2_));
-      if (pay_bytes > half)
-        HIP_CHECK(hipMemcpyAsync(d_stage_pay_[slot] + half,
-                                 reinterpret_cast<u8 *>(pay_ptr) + half,
-                                 pay_bytes - half, hipMemcpyHostToDevice,
-                                 h2d_stream_));
+                                 reinterpret_cast<void *>(pay_ptr), pay_bytes,
+                                 hipMemcpyHostToDevice, h2d_stream_));
       HIP_CHECK(hipEventRecord(stage_ev_[slot], h2d_stream_));
       HIP_CHECK(hipEventRecord(up_ev_[slot], h2d_stream_));
-      HIP_CHECK(hipEventRecord(up2_ev_[slot], h2d_stream2_));
     }
     staged_n_[slot] = n;
     staged_pay_[slot] = pay_bytes;
@@ -1032,9 +1016,8 @@ public:
     {
       py::gil_scoped_release nogil;
       if (uploaded_[slot]) {
-        // already uploaded by prefetch: order kernels after both halves
+        // already uploaded by prefetch_staged: order kernels after it
         HIP_CHECK(hipStreamWaitEvent(stream_, up_ev_[slot], 0));
-        HIP_CHECK(hipStreamWaitEvent(stream_, up2_ev_[slot], 0));
         uploaded_[slot] = false;
       } else {
         HIP_CHECK(hipMemcpyAsync(d_stage_recs_[slot], h_recs_[slot],
@@ -1123,7 +1106,6 @@ public:
       py::gil_scoped_release nogil;
       if (uploaded_[slot]) {
         HIP_CHECK(hipStreamWaitEvent(stream_, up_ev_[slot], 0));
-        HIP_CHECK(hipStreamWaitEvent(stream_, up2_ev_[slot], 0));
         uploaded_[slot] = false;
       }
       HIP_CHECK(hipGraphLaunch(tick_exec_[slot], stream_));
@@ -1646,7 +1628,6 @@ private:
     // all three streams may hold work against the staging buffers
     HIP_CHECK(hipStreamSynchronize(stream_));
     HIP_CHECK(hipStreamSynchronize(h2d_stream_));
-    HIP_CHECK(hipStreamSynchronize(h2d_stream2_));
     HIP_CHECK(hipStreamSynchronize(copy_stream_));
     for (int s = 0; s < 2; ++s) {
       HIP_CHECK(hipFree(d_stage_pay_[s]));
@@ -1669,8 +1650,7 @@ private:
   size_t out_pool_ = 0;
   size_t stage_pay_bytes_ = 0;
 
-  hipStream_t stream_{}, copy_stream_{}, h2d_stream_{}, h2d_stream2_{};
-  hipEvent_t up2_ev_[2] = {};
+  hipStream_t stream_{}, copy_stream_{}, h2d_stream_{};
   hipEvent_t ev_{};
   hipEvent_t up_ev_[2] = {};
   hipEvent_t d2h_ev_[2] = {};
