@@ -457,3 +457,20 @@ def test_receive_timeout_waits_for_message(tmp_db):
     start = _t.monotonic()
     assert tmp_db.receive_messages("waiter", timeout=0.1) == []
     assert _t.monotonic() - start >= 0.09
+
+
+def test_derived_id_from_other_rank_not_found(tmp_db):
+    from swarmdb_amd.core.wire import derived_id
+
+    tmp_db.send_message("a", "x", receiver_id="b")
+    # rank 7's derived id never resolves on rank 0
+    assert tmp_db.get_message(derived_id(7, 0)) is None
+    # out-of-range seq on our rank doesn't resolve either
+    assert tmp_db.get_message(derived_id(0, 999999)) is None
+
+
+def test_unregistered_agent_receive_empty(tmp_db):
+    # engine-level receive for an index that never registered
+    assert len(tmp_db.engine.receive(37, 10)) == 0
+    assert tmp_db.engine.unread_count(37) == 0
+    assert len(tmp_db.engine.peek_inbox(37)) == 0
