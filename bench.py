@@ -293,7 +293,11 @@ def main() -> int:
             if not args.no_gather and ndel:
                 # device-side gather straight from the dequeue output
                 # buffer; D2H overlaps the next tick (bounded 2 deep)
-                q.deliver_outbuf(ndel, args.payload, False)
+                c64 = counts.astype(np.int64)
+                offs = np.zeros(len(c64), dtype=np.uint32)
+                offs[1:] = np.cumsum(c64[:-1]).astype(np.uint32)
+                q.deliver_outbuf(offs, ndel, int(c64.max()),
+                                 args.payload, False)
             sent_total += n_staged
             recv_total += ndel
             _cur[0] = 1 - cur

@@ -715,7 +715,9 @@ def test_deliver_outbuf_bytes(gpu_engine):
     counts, seqs = eng.q.receive_many(agents, 100, False, return_seqs=True)
     total = int(counts.sum())
     assert total == n
-    nbytes = eng.q.deliver_outbuf(total, plen, True)
+    offs = np.zeros(n_agents, dtype=np.uint32)
+    offs[1:] = np.cumsum(counts.astype(np.int64)[:-1]).astype(np.uint32)
+    nbytes = eng.q.deliver_outbuf(offs, total, int(counts.max()), plen, True)
     assert nbytes == total * plen
     # cross-check against fetch() of the same seqs in delivery order
     order = []
