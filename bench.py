@@ -229,8 +229,9 @@ def main() -> int:
 
     # single-GPU pipelined path: batches live in PINNED host buffers and
     # upload on a dedicated H2D stream, overlapping the previous tick's
-    # kernels and delivery D2H — zero host copies on the send path
-    pipelined = have_gpu and router is None and hasattr(engine, "q")
+    # kernels and delivery D2H — zero host copies on the send path.
+    # (Never in distributed mode: the routed step above owns that case.)
+    pipelined = have_gpu and not dist_on and hasattr(engine, "q")
     if pipelined:
         q = engine.q
         pinned = []
