@@ -339,6 +339,15 @@ def main() -> int:
     if args.dump_steps and rank == 0:
         print("step_times_ms:", [round(t * 1000, 3) for t in step_times])
 
+    # conservation: at steady state every enqueued message is delivered
+    # within the same tick; a big gap means routing or dequeue is broken
+    if sent_total and abs(recv_total - sent_total) > 0.05 * sent_total:
+        print(
+            f"[bench] WARNING rank {rank}: delivered {recv_total} of "
+            f"{sent_total} sent — routing/dequeue imbalance",
+            file=sys.stderr,
+        )
+
     msgs_per_s = recv_all / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
     p50_ms = float(np.median(step_times) * 1000.0)
