@@ -133,6 +133,19 @@ def main() -> int:
         # HBM writes and less PCIe on the gather than a fixed 2 KiB slot
         slot_bytes=max(256, ((args.payload + 15) // 16) * 16 + 64),
         inbox_capacity=1 << 16,
+        # dequeue window: 4x the expected poll depth (pow2, min 256);
+        # small windows keep many dequeue workgroups resident
+        recv_window=min(
+            4096,
+            max(
+                256,
+                1
+                << (
+                    4 * max(64, 4 * args.batch * world // max(1, args.agents))
+                    - 1
+                ).bit_length(),
+            ),
+        ),
         staging_batch=max(16384, args.batch * (2 if dist_on else 1)),
         device_index=local_rank % max(1, n_dev),
         auto_save=False,

@@ -240,8 +240,13 @@ __global__ void __launch_bounds__(256)
               const u64 *__restrict__ bitmaps, u64 *__restrict__ out_seqs,
               u32 *__restrict__ out_counts, ull *__restrict__ by_status,
               ull *__restrict__ received, QueueGeom g) {
-  __shared__ u64 keys[RECV_WINDOW]; // 32 KiB of 160 KiB LDS
-  __shared__ u32 sh_valid;
+  // dynamic LDS: [0..1] control words, [2..] the sort window (the
+  // window size is a queue parameter — small dequeue windows keep many
+  // blocks resident; the base stays 16-B aligned with no static
+  // __shared__ objects, guide §6 G17)
+  extern __shared__ __attribute__((aligned(16))) u64 smem[];
+  u64 *keys = smem + 2;
+  u32 *sh_valid = reinterpret_cast<u32 *>(smem);
 
   const int b = blockIdx.x;
   if (b >= n_agents)
@@ -318,10 +323,10 @@ __global__ void __launch_bounds__(256)
       else
         lo = mid + 1;
     }
-    sh_valid = lo;
+    *sh_valid = lo;
   }
   __syncthreads();
-  const u32 valid = sh_valid;
+  const u32 valid = *sh_valid;
   const u32 take = valid < (u32)max_per_agent ? valid : (u32)max_per_agent;
 
   for (u32 i = threadIdx.x; i < take; i += blockDim.x) {
@@ -687,7 +692,7 @@ class DeviceQueue {
 public:
   DeviceQueue(u32 num_slots, u32 slot_bytes, u32 max_agents,
               u32 inbox_capacity, u32 num_bitmaps, u32 num_backends,
-              u32 staging_batch, int device)
+              u32 staging_batch, int device, u32 recv_window = RECV_WINDOW)
       : device_(device), staging_batch_(staging_batch) {
     if (slot_bytes % 16 != 0)
       throw std::invalid_argument("slot_bytes must be a multiple of 16");
@@ -700,7 +705,9 @@ public:
     g_.num_bitmaps = num_bitmaps;
     g_.bitmap_words = max_agents / 64;
     g_.num_backends = num_backends;
-    g_.recv_window = RECV_WINDOW;
+    if (recv_window < 64 || (recv_window & (recv_window - 1)))
+      throw std::invalid_argument("recv_window must be a power of two >= 64");
+    g_.recv_window = recv_window;
 
     HIP_CHECK(hipSetDevice(device_));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
@@ -720,8 +727,8 @@ public:
         hipMalloc(&d_inbox_, (size_t)max_agents * inbox_capacity * sizeof(u64)));
     HIP_CHECK(hipMalloc(&d_wpos_, (size_t)max_agents * sizeof(ull)));
     HIP_CHECK(hipMalloc(&d_rpos_, (size_t)max_agents * sizeof(ull)));
-    HIP_CHECK(
-        hipMalloc(&d_carry_, (size_t)max_agents * RECV_WINDOW * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_carry_,
+                        (size_t)max_agents * g_.recv_window * sizeof(u64)));
     HIP_CHECK(hipMalloc(&d_carry_n_, (size_t)max_agents * sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_active_, (size_t)max_agents * sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_bitmaps_,
@@ -1062,7 +1069,8 @@ public:
                          dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
                          d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
                          g_);
-      hipLaunchKernelGGL(k_receive, dim3(na), dim3(256), 0, stream_,
+      hipLaunchKernelGGL(k_receive, dim3(na), dim3(256),
+                         (g_.recv_window + 2) * sizeof(u64), stream_,
                          d_agents_, na, K, priority ? 1 : 0, 0, d_dyn_,
                          d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
                          d_carry_, d_carry_n_, d_bitmaps_, d_out_seqs_,
@@ -1193,7 +1201,9 @@ public:
       py::gil_scoped_release nogil;
       HIP_CHECK(hipMemcpyAsync(d_agents_, agents.data(), na * sizeof(u32),
                                hipMemcpyHostToDevice, stream_));
-      hipLaunchKernelGGL(k_receive, dim3(na), dim3(256), 0, stream_, d_agents_,
+      hipLaunchKernelGGL(k_receive, dim3(na), dim3(256),
+                         (g_.recv_window + 2) * sizeof(u64), stream_,
+                         d_agents_,
                          na, max_per_agent, priority ? 1 : 0, evict_base_,
                          (const u64 *)nullptr,
                          d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
@@ -1716,11 +1726,11 @@ PYBIND11_MODULE(_swarmq, m) {
   });
 
   py::class_<DeviceQueue>(m, "DeviceQueue")
-      .def(py::init<u32, u32, u32, u32, u32, u32, u32, int>(),
+      .def(py::init<u32, u32, u32, u32, u32, u32, u32, int, u32>(),
            py::arg("num_slots"), py::arg("slot_bytes"), py::arg("max_agents"),
            py::arg("inbox_capacity"), py::arg("num_bitmaps"),
            py::arg("num_backends"), py::arg("staging_batch"),
-           py::arg("device") = 0)
+           py::arg("device") = 0, py::arg("recv_window") = RECV_WINDOW)
       .def("register_agent", &DeviceQueue::register_agent)
       .def("deregister_agent", &DeviceQueue::deregister_agent)
       .def("active_agents", &DeviceQueue::active_agents)

@@ -38,6 +38,9 @@ class QueueConfig:
     slot_bytes: int = 2048               # fixed message slot (header+payload)
     num_slots: int = 1 << 20             # ring capacity in slots (2 GiB @2KB)
     inbox_capacity: int = 1 << 16        # per-agent inbox ring entries (u64)
+    recv_window: int = 4096              # entries examined per dequeue call
+    #                                      (pow2; small windows keep many
+    #                                      dequeue workgroups resident)
     staging_batch: int = 16384           # max messages per enqueue batch
     num_backends: int = 64               # LLM backend table capacity
     use_gpu: Optional[bool] = None       # None = auto-detect

@@ -87,7 +87,7 @@ class CpuEngine(Engine):
         # ring itself is append-only history; the GPU engine keeps the
         # same per-agent carry buffer device-side)
         self._carry: Dict[int, np.ndarray] = {}
-        self._window = 4096  # max entries examined per receive call
+        self._window = c.recv_window  # max entries examined per receive
 
         self._bitmaps: List[np.ndarray] = []
 

@@ -64,6 +64,7 @@ class GpuEngine(Engine):
             num_backends=c.num_backends,
             staging_batch=c.staging_batch,
             device=c.device_index,
+            recv_window=c.recv_window,
         )
         self._staging = c.staging_batch
         self._slot_bytes = c.slot_bytes
