@@ -208,6 +208,18 @@ class DistributedSwarmsDB(SwarmsDB):
             token_count=self._count_tokens(content),
             visible_to=vis,
         )
+        # the single-process facade spills oversized content to a host
+        # overflow store; across ranks the payload must fit a device slot
+        # (the owner rank has no host copy) — fail loudly at send time
+        approx = len(encode_content(content)[0]) + len(
+            encode_extras(msg.id, msg.metadata, vis)
+        )
+        if approx > int(self.config.slot_bytes):
+            raise ValueError(
+                f"content ({approx} B) exceeds the distributed slot "
+                f"capacity ({self.config.slot_bytes} B); raise "
+                "SWARMQ_SLOT_BYTES"
+            )
         with self._lock:
             self._out_msgs.append(msg)
             self._pending_meta[msg.id] = msg
