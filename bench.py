@@ -85,6 +85,9 @@ def main() -> int:
                     help="payload bytes (1 KB chat messages)")
     ap.add_argument("--priority", action="store_true",
                     help="priority-ordered dequeue (config 3 kernel)")
+    ap.add_argument("--bcast-frac", type=float, default=0.0,
+                    help="fraction of messages sent as broadcasts "
+                         "(each fans out to every agent)")
     ap.add_argument("--no-gather", action="store_true",
                     help="skip payload D2H gather (delivery stays device-side)")
     ap.add_argument("--dump-steps", action="store_true",
@@ -163,7 +166,7 @@ def main() -> int:
 
     batches, local_agents = build_batches(
         rng, min(8, max(2, args.steps)), args.batch, agents_global, rank,
-        world, args.payload,
+        world, args.payload, bcast_frac=args.bcast_frac,
     )
     for a in local_agents:
         engine.register_agent(int(a))
@@ -183,7 +186,9 @@ def main() -> int:
 
             router = CrossGpuRouter(torch.device("cpu"))
 
-    recv_K = max(64, 4 * args.batch * world // max(1, len(local_agents)))
+    per_agent = args.batch * world / max(1, agents_global)
+    per_agent += args.bcast_frac * args.batch * world  # every bcast hits all
+    recv_K = max(64, int(4 * per_agent))
     sent_total = 0
     recv_total = 0
 
