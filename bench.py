@@ -304,19 +304,30 @@ def main() -> int:
                 # upload batch i+1 on the H2D stream: overlaps batch i's
                 # kernels and the delivery D2H (full-duplex PCIe)
                 _prefetch(1 - cur, i + 1)
-                counts, _ = q.receive_many(
+                want_seqs = args.bcast_frac > 0
+                counts, seqs = q.receive_many(
                     local_agents32, recv_K, bool(args.priority),
-                    return_seqs=False,
+                    return_seqs=want_seqs,
                 )
                 ndel = int(counts.astype(np.int64).sum())
             if not args.no_gather and ndel:
-                # device-side gather straight from the dequeue output
-                # buffer; D2H overlaps the next tick (bounded 2 deep)
-                c64 = counts.astype(np.int64)
-                offs = np.zeros(len(c64), dtype=np.uint32)
-                offs[1:] = np.cumsum(c64[:-1]).astype(np.uint32)
-                q.deliver_outbuf(offs, ndel, int(c64.max()),
-                                 args.payload, False)
+                stride16 = (args.payload + 15) // 16 * 16
+                fits = ndel * stride16 <= engine.cfg.staging_batch * engine.cfg.slot_bytes
+                if fits:
+                    # device-side gather straight from the dequeue output
+                    # buffer; D2H overlaps the next tick (bounded 2 deep)
+                    c64 = counts.astype(np.int64)
+                    offs = np.zeros(len(c64), dtype=np.uint32)
+                    offs[1:] = np.cumsum(c64[:-1]).astype(np.uint32)
+                    q.deliver_outbuf(offs, ndel, int(c64.max()),
+                                     args.payload, False)
+                else:
+                    # fan-out-heavy tick overflows the pinned staging:
+                    # chunked delivery via the seq path
+                    c64 = counts.astype(np.int64)
+                    mat = seqs.reshape(len(local_agents32), recv_K)
+                    taken = mat[np.arange(recv_K)[None, :] < c64[:, None]]
+                    engine.deliver_payloads(taken, args.payload)
             sent_total += n_staged
             recv_total += ndel
             _cur[0] = 1 - cur
