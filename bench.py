@@ -277,7 +277,7 @@ def main() -> int:
 
         # hipGraph-captured steady-state tick: enqueue + fanout +
         # receive + result D2H replay as ONE graph launch per step
-        use_graph = args.graph
+        use_graph = args.graph and args.bcast_frac == 0
         if use_graph:
             try:
                 q.build_tick(args.batch, local_agents.astype(np.uint32),
