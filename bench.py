@@ -369,11 +369,13 @@ def main() -> int:
         print("step_times_ms:", [round(t * 1000, 3) for t in step_times])
 
     # conservation: at steady state every enqueued message is delivered
-    # within the same tick; a big gap means routing or dequeue is broken
-    if sent_total and abs(recv_total - sent_total) > 0.05 * sent_total:
+    # within the same tick (broadcasts amplify, so only LOSS warns;
+    # p2p-only runs should match within 5%)
+    expect = sent_total * (1 + args.bcast_frac * max(0, args.agents - 1))
+    if sent_total and recv_total < 0.95 * expect:
         print(
             f"[bench] WARNING rank {rank}: delivered {recv_total} of "
-            f"{sent_total} sent — routing/dequeue imbalance",
+            f"~{int(expect)} expected — routing/dequeue loss",
             file=sys.stderr,
         )
 
