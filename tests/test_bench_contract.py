@@ -63,3 +63,16 @@ def test_bench_torchrun_world2():
     assert out["config"]["global_batch"] == 512
     assert "all-to-all" in out["config"]["parallelism"]
     assert out["value"] > 0
+
+
+def test_bench_broadcast_mode():
+    proc = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "3", "--warmup", "1",
+         "--batch", "256", "--agents", "64", "--bcast-frac", "0.05"],
+        capture_output=True, text=True, timeout=300, cwd=str(REPO),
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    out = _last_json_line(proc.stdout)
+    assert out["value"] > 0
+    # fan-out amplifies deliveries; no loss warning expected
+    assert "loss" not in proc.stderr
