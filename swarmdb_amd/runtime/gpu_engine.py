@@ -121,8 +121,12 @@ class GpuEngine(Engine):
             while done < n:
                 chunk = min(self._staging, n - done)
                 sub = recs[done : done + chunk]
-                lo = int(sub["payload_off"][0])
-                hi = int(sub["payload_off"][-1] + sub["payload_len"][-1])
+                # offsets need not be monotonic: bound the slice by the
+                # actual extremes
+                lo = int(sub["payload_off"].min())
+                hi = int(
+                    (sub["payload_off"] + sub["payload_len"]).max()
+                )
                 if done or lo:
                     sub = sub.copy()
                     sub["payload_off"] -= np.uint64(lo)
