@@ -226,6 +226,21 @@ def create_app(
                 status_code=http.HTTP_401_UNAUTHORIZED,
                 detail="Invalid credentials",
             )
+        if credential_validator is None and settings.api_env not in (
+            "development",
+            "dev",
+            "test",
+        ) and is_admin(credentials.username):
+            # open demo auth must not self-mint admin outside development:
+            # without a validator any peer could claim sub == "admin" and
+            # reach the admin routes (incl. /admin/load's file ingest)
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail=(
+                    "admin login requires a credential validator outside "
+                    "development mode"
+                ),
+            )
         return Token(access_token=create_access_token(credentials.username))
 
     # ---------------- agents ----------------
@@ -333,6 +348,24 @@ def create_app(
 
         if not messages:
             return {"status": "sent", "message_ids": []}
+        if any(m.visible_to or m.metadata for m in messages):
+            # restricted/metadata-bearing items need the extras+bitmap
+            # path (the fast path below would deliver a visible_to-
+            # restricted message to everyone); still ONE engine batch
+            msgs = [
+                db.make_message(
+                    sender_id=current,
+                    content=m.content,
+                    receiver_id=m.receiver_id,
+                    message_type=m.message_type,
+                    priority=m.priority,
+                    metadata=m.metadata,
+                    visible_to=m.visible_to,
+                )
+                for m in messages
+            ]
+            ids = db.send_messages_bulk(msgs)
+            return {"status": "sent", "message_ids": ids}
         sidx = db.agent_index(current)
         n = len(messages)
         recs = np.zeros(n, dtype=REC_DTYPE)
@@ -714,10 +747,21 @@ def create_app(
         path: str = Query(...), current: str = Depends(get_current_agent)
     ):
         """load_message_history over HTTP (never routed in the reference —
-        SURVEY.md §5.4)."""
+        SURVEY.md §5.4). The path is constrained to files under the
+        configured save_dir: an admin token must not turn the server into
+        an arbitrary-filesystem JSON reader."""
         _require_admin(current)
+        from pathlib import Path as _Path
+
+        save_root = _Path(db.save_dir).resolve()
+        target = _Path(path).resolve()
+        if save_root not in target.parents and target != save_root:
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="history path must live under the save directory",
+            )
         try:
-            n = db.load_message_history(path)
+            n = db.load_message_history(target)
         except FileNotFoundError:
             raise HTTPException(
                 status_code=http.HTTP_404_NOT_FOUND,

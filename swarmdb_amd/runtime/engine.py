@@ -227,6 +227,11 @@ class Engine(abc.ABC):
     @abc.abstractmethod
     def total_messages(self) -> int: ...
 
+    def evict_base(self) -> int:
+        """Lowest seq still retained (slot-ring retention horizon). 0 on
+        engines that never evict (the CPU log is append-only)."""
+        return 0
+
     @abc.abstractmethod
     def stats_arrays(self) -> Dict[str, np.ndarray]:
         """Running counters: by_type [7], by_status [6], sent [max_agents],

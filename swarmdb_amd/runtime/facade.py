@@ -85,6 +85,8 @@ class SwarmsDB:
     behavior mirrors.
     """
 
+    _PRUNE_EVERY = 4096  # compat-map sweep cadence (inserts)
+
     def __init__(
         self,
         config: Optional[QueueConfig] = None,
@@ -137,6 +139,8 @@ class SwarmsDB:
         # seq -> (content_len, payload bytes)
         # (SURVEY.md §7 'variable-size content in fixed slots')
         self._overflow: Dict[int, tuple] = {}
+        self._prune_tick = 0
+        self._pruned_below = 0
 
         self._llm_backends: List[str] = []
         self._llm_backend_idx: Dict[str, int] = {}
@@ -352,6 +356,29 @@ class SwarmsDB:
         self._id_to_seq[msg.id] = seq
         if overflow is not None:
             self._overflow[seq] = overflow
+        self._maybe_prune_host_maps()
+
+    def _maybe_prune_host_maps(self) -> None:
+        """Evict ``_id_to_seq``/``_overflow`` entries whose seq fell below
+        the device retention horizon (the slot ring's evict_base) — the
+        compat-path host maps must not outlive the messages they index,
+        or a long-running service leaks host RAM while the ring evicts.
+        Amortized: a full sweep every ``_PRUNE_EVERY`` inserts."""
+        self._prune_tick += 1
+        if self._prune_tick < self._PRUNE_EVERY:
+            return
+        self._prune_tick = 0
+        eb = self.engine.evict_base()
+        if eb <= self._pruned_below:
+            return
+        self._id_to_seq = {
+            k: v for k, v in self._id_to_seq.items() if v >= eb
+        }
+        if self._overflow:
+            self._overflow = {
+                s: blob for s, blob in self._overflow.items() if s >= eb
+            }
+        self._pruned_below = eb
 
     def send_messages_bulk(self, msgs: List[Message]) -> List[str]:
         """Enqueue many validated Messages as ONE engine batch (the
@@ -692,11 +719,14 @@ class SwarmsDB:
             if idx is None:
                 return []
         seqs = self.engine.peek_inbox(idx)[::-1]  # newest first
+        # reference pagination order (main.py:640-652): skip raw inbox
+        # entries first, THEN apply the status filter, THEN cap at limit
+        seqs = seqs[skip:]
         if status is not None and len(seqs):
             code = _status_code(status)
             st = self.engine.statuses(seqs)
             seqs = seqs[st == code]
-        seqs = seqs[skip : skip + limit]
+        seqs = seqs[:limit]
         return self._messages_from_seqs(seqs.copy())
 
     def mark_message_as_processed(self, message_id: str) -> bool:

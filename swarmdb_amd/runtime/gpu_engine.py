@@ -69,8 +69,9 @@ class GpuEngine(Engine):
         self._staging = c.staging_batch
         self._slot_bytes = c.slot_bytes
         # host-side receive-event log (processing_rate probe); one entry
-        # per poll tick, pruned as it grows
+        # per poll tick, pruned by age (see receive_many)
         self._recv_events: List[Tuple[float, np.ndarray, np.ndarray]] = []
+        self._RECV_EVENT_RETAIN_S = 600.0  # max supported probe window
 
     # --- registry ---
 
@@ -175,10 +176,20 @@ class GpuEngine(Engine):
         mat = flat.reshape(len(agent_idxs), max_per_agent)
         mask = np.arange(max_per_agent)[None, :] < counts[:, None]
         seqs = mat[mask]
-        # O(1)-per-tick receive-rate bookkeeping (processing_rate probe)
-        self._recv_events.append((time.time(), agent_idxs, counts))
-        if len(self._recv_events) > 1024:
-            del self._recv_events[:512]
+        # O(1)-per-tick receive-rate bookkeeping (processing_rate probe).
+        # Prune by AGE, not entry count: a fixed-count prune truncated the
+        # 60 s probe window under sustained polling (>~17 ticks/s). The
+        # retention bound is _RECV_EVENT_RETAIN_S — the largest window
+        # recv_rate_window supports exactly.
+        now = time.time()
+        self._recv_events.append((now, agent_idxs, counts))
+        cutoff = now - self._RECV_EVENT_RETAIN_S
+        if self._recv_events[0][0] < cutoff:
+            i = 0
+            for i, (t, _, _) in enumerate(self._recv_events):
+                if t >= cutoff:
+                    break
+            del self._recv_events[:i]
         return counts, seqs
 
     def peek_inbox(self, agent_idx: int) -> np.ndarray:
@@ -379,6 +390,9 @@ class GpuEngine(Engine):
 
     def total_messages(self) -> int:
         return int(self.q.total_messages())
+
+    def evict_base(self) -> int:
+        return int(self.q.evict_base())
 
     def stats_arrays(self) -> Dict[str, np.ndarray]:
         c = self.q.counters()

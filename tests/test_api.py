@@ -280,7 +280,12 @@ def test_admin_load_endpoint(client):
     path = client.post("/admin/save", headers=ha).json()["path"]
     r = client.post(f"/admin/load?path={path}", headers=ha)
     assert r.status_code == 200
-    r = client.post("/admin/load?path=/nonexistent.json", headers=ha)
+    # paths outside the save dir are rejected (no arbitrary-file ingest)
+    r = client.post("/admin/load?path=/etc/passwd", headers=ha)
+    assert r.status_code == 403
+    # missing files *under* the save dir are a plain 404
+    missing = str(client.db.save_dir / "nonexistent.json")
+    r = client.post(f"/admin/load?path={missing}", headers=ha)
     assert r.status_code == 404
 
 
@@ -439,3 +444,91 @@ def test_micro_batched_sends(tmp_path):
                      headers=h2).json()
         assert {m["id"] for m in got} == ids
     db.config.auto_save = False
+
+
+def test_batch_send_respects_visible_to(client):
+    """A batch item restricted via visible_to must not reach agents
+    outside the list (round-1 advisor finding: the fast path hardcoded
+    VIS_ALL and leaked restricted broadcasts to everyone)."""
+    ha = auth(client, "alice")
+    hb = auth(client, "bob")
+    he = auth(client, "eve")
+    for name, h in [("bob", hb), ("eve", he)]:
+        client.post("/agents/register", headers=h, json={"agent_id": name})
+    msgs = [
+        {"receiver_id": None, "content": "secret", "visible_to": ["bob"]},
+        {"receiver_id": "bob", "content": "tagged",
+         "metadata": {"k": "v"}},
+    ]
+    r = client.post("/messages/batch", headers=ha, json=msgs)
+    assert r.status_code == 200, r.text
+    got_bob = client.post(
+        "/agents/receive?timeout=0&max_messages=100", headers=hb
+    ).json()
+    got_eve = client.post(
+        "/agents/receive?timeout=0&max_messages=100", headers=he
+    ).json()
+    assert sorted(m["content"] for m in got_bob) == ["secret", "tagged"]
+    assert [m["content"] for m in got_eve] == []  # eve never sees it
+    tagged = [m for m in got_bob if m["content"] == "tagged"][0]
+    assert tagged["metadata"] == {"k": "v"}
+
+
+def test_admin_login_blocked_outside_development(tmp_path, monkeypatch):
+    """Open demo auth must not mint admin outside development mode when
+    no credential validator is installed (round-1 advisor finding)."""
+    monkeypatch.setenv("API_ENV", "production")
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path / "hist"),
+                      max_agents=64, auto_save=False)
+    db = SwarmsDB(config=cfg)
+    app = create_app(db=db, settings=ApiSettings())
+    with TestClient(app) as c:
+        r = c.post("/auth/token",
+                   json={"username": "admin", "password": "pw"})
+        assert r.status_code == 403
+        # non-admin demo logins still work (documented demo behavior)
+        r = c.post("/auth/token",
+                   json={"username": "alice", "password": "pw"})
+        assert r.status_code == 200
+    # with a validator, admin can authenticate in production
+    app2 = create_app(
+        db=db,
+        settings=ApiSettings(),
+        credential_validator=lambda u, p: p == "s3cret",
+    )
+    with TestClient(app2) as c:
+        r = c.post("/auth/token",
+                   json={"username": "admin", "password": "s3cret"})
+        assert r.status_code == 200
+        r = c.post("/auth/token",
+                   json={"username": "admin", "password": "wrong"})
+        assert r.status_code == 401
+
+
+def test_agent_messages_pagination_matches_reference_order(client):
+    """skip applies to raw newest-first inbox entries BEFORE the status
+    filter (reference main.py:640-652)."""
+    ha, hb = auth(client, "a"), auth(client, "b")
+    for i in range(5):
+        client.post("/messages", headers=ha,
+                    json={"receiver_id": "b", "content": f"m{i}"})
+    # read everything, then mark m4 (newest) processed
+    got = client.post("/agents/receive?timeout=0&max_messages=100",
+                      headers=hb).json()
+    assert len(got) == 5
+    newest = [m for m in got if m["content"] == "m4"][0]
+    client.put(f"/messages/{newest['id']}/status?status=processed",
+               headers=hb)
+    # newest-first raw order: m4(processed), m3..m0(read)
+    # skip=1 drops m4 BEFORE filtering; status=read then yields m3..m0
+    r = client.get("/agents/b/messages?status=read&skip=1&limit=10",
+                   headers=hb)
+    assert [m["content"] for m in r.json()] == ["m3", "m2", "m1", "m0"]
+    # skip=0 with status=processed yields only m4
+    r = client.get("/agents/b/messages?status=processed&skip=0&limit=10",
+                   headers=hb)
+    assert [m["content"] for m in r.json()] == ["m4"]
+    # skip=1 with status=processed: m4 is skipped as a RAW entry -> empty
+    r = client.get("/agents/b/messages?status=processed&skip=1&limit=10",
+                   headers=hb)
+    assert r.json() == []

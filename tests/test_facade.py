@@ -474,3 +474,25 @@ def test_unregistered_agent_receive_empty(tmp_db):
     assert len(tmp_db.engine.receive(37, 10)) == 0
     assert tmp_db.engine.unread_count(37) == 0
     assert len(tmp_db.engine.peek_inbox(37)) == 0
+
+
+def test_compat_host_maps_prune_with_retention(tmp_path):
+    """_id_to_seq/_overflow must not outlive the engine's retention
+    horizon (round-1 review: unbounded host maps leak RAM while the
+    device ring evicts)."""
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path), auto_save=False,
+                      max_agents=64)
+    db = SwarmsDB(config=cfg)
+    db._PRUNE_EVERY = 8  # tight cadence for the test
+    ids = [db.send_message("a", f"m{i}", receiver_id="b") for i in range(32)]
+    assert len(db._id_to_seq) == 32
+    # simulate the device ring evicting the first 16 seqs
+    db.engine.evict_base = lambda: 16
+    for i in range(db._PRUNE_EVERY + 1):
+        db.send_message("a", f"late{i}", receiver_id="b")
+    live = set(db._id_to_seq.values())
+    assert all(s >= 16 for s in live)
+    # ids below the horizon are gone; newer ones survive
+    assert ids[0] not in db._id_to_seq
+    assert db._id_to_seq.get(ids[-1]) is not None
+    db.close()
