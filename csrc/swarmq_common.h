@@ -22,6 +22,7 @@ using i64 = int64_t;
 // ---- codes (must match swarmdb_amd/runtime/engine.py) ----
 constexpr u32 BROADCAST = 0xFFFFFFFFu;
 constexpr u32 NO_BITMAP = 0xFFFFFFFFu;
+constexpr u32 EPOCH_NONE = 0xFFFFFFFFu; // bitmap-epoch slot "never allocated"
 
 constexpr u32 ST_PENDING = 0;
 constexpr u32 ST_DELIVERED = 1;
@@ -51,7 +52,10 @@ struct __attribute__((aligned(16))) Rec {
   u32 payload_len;
   u32 bitmap;       // visibility bitmap index or NO_BITMAP
   u32 content_len;  // content prefix of the payload (search window)
-  u32 reserved;
+  u32 bitmap_epoch; // allocation counter of the referenced bitmap; the
+                    // pool is a ring, so dequeue/fan-out verify
+                    // bitmap_epochs[bitmap] == bitmap_epoch and treat a
+                    // recycled entry as NOT visible (never misdelivered)
 };
 static_assert(sizeof(Rec) == 48, "Rec layout must match REC_DTYPE");
 

@@ -23,6 +23,8 @@
 #include <pybind11/stl.h>
 
 #include <algorithm>
+#include <atomic>
+#include <chrono>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -99,11 +101,13 @@ __global__ void k_enqueue_meta(const Rec *__restrict__ stage, int n,
                                u32 *__restrict__ status,
                                u64 *__restrict__ inbox,
                                ull *__restrict__ inbox_wpos,
+                               const ull *__restrict__ inbox_rpos,
                                ull *__restrict__ by_type,
                                ull *__restrict__ by_status,
                                ull *__restrict__ sent,
                                u64 *__restrict__ bcast_list,
-                               u32 *__restrict__ bcast_count, QueueGeom g) {
+                               u32 *__restrict__ bcast_count,
+                               ull *__restrict__ dropped, QueueGeom g) {
   // per-block histograms: one global atomic per bin per block
   __shared__ u32 h_type[N_TYPES];
   __shared__ u32 h_msgs;
@@ -150,6 +154,12 @@ __global__ void k_enqueue_meta(const Rec *__restrict__ stage, int n,
         bcast_list[bi] = seq;
       } else {
         ull pos = atomicAdd(&inbox_wpos[r.receiver], 1ull);
+        // ring overfill: a slow consumer's unexamined entry is about to
+        // be overwritten — count the loss instead of hiding it (exposed
+        // via stats; the CPU engine's inboxes are unbounded, so this is
+        // the one place GPU delivery can drop)
+        if (pos - inbox_rpos[r.receiver] >= (ull)g.inbox_capacity)
+          atomicAdd(dropped, 1ull);
         inbox[(u64)r.receiver * g.inbox_capacity + (pos % g.inbox_capacity)] =
             seq;
       }
@@ -196,8 +206,10 @@ __global__ void k_fanout(const u64 *__restrict__ bcast_list,
                          const u32 *__restrict__ active,
                          const Rec *__restrict__ hdr,
                          const u64 *__restrict__ bitmaps,
+                         const u32 *__restrict__ bitmap_epochs,
                          u64 *__restrict__ inbox, ull *__restrict__ inbox_wpos,
-                         QueueGeom g) {
+                         const ull *__restrict__ inbox_rpos,
+                         ull *__restrict__ dropped, QueueGeom g) {
   const u32 a = blockIdx.x * blockDim.x + threadIdx.x;
   if (a >= g.max_agents || !active[a])
     return;
@@ -205,21 +217,30 @@ __global__ void k_fanout(const u64 *__restrict__ bcast_list,
   if (nb == 0)
     return;
   ull pos = inbox_wpos[a];
+  const ull rpos = inbox_rpos[a];
+  ull ndrop = 0;
   u64 *ib = inbox + (u64)a * g.inbox_capacity;
   for (u32 i = 0; i < nb; ++i) {
     const u64 seq = bcast_list[i];
     const Rec &h = hdr[seq % g.num_slots]; // uniform per i, L2-cached
     if (h.vis_mode == VIS_GROUP) {
-      // group fan-out: only member inboxes get the entry
+      // group fan-out: only member inboxes get the entry; a recycled
+      // bitmap (epoch mismatch) means membership is unknowable — skip
+      // rather than misdeliver
       if (h.bitmap == NO_BITMAP ||
+          bitmap_epochs[h.bitmap] != h.bitmap_epoch ||
           !((bitmaps[(u64)h.bitmap * g.bitmap_words + (a >> 6)] >>
              (a & 63)) & 1ull))
         continue;
     }
+    if (pos - rpos >= (ull)g.inbox_capacity)
+      ++ndrop; // ring overfill overwrites an unexamined entry
     ib[pos % g.inbox_capacity] = seq;
     ++pos;
   }
   inbox_wpos[a] = pos;
+  if (ndrop)
+    atomicAdd(dropped, ndrop);
 }
 
 // Dequeue: one 256-thread workgroup per polling agent. Drains the
@@ -237,7 +258,9 @@ __global__ void __launch_bounds__(256)
               u32 *__restrict__ status, const u64 *__restrict__ inbox,
               ull *__restrict__ inbox_wpos, ull *__restrict__ inbox_rpos,
               u64 *__restrict__ carry, u32 *__restrict__ carry_n,
-              const u64 *__restrict__ bitmaps, u64 *__restrict__ out_seqs,
+              const u64 *__restrict__ bitmaps,
+              const u32 *__restrict__ bitmap_epochs,
+              u64 *__restrict__ out_seqs,
               u32 *__restrict__ out_counts, ull *__restrict__ by_status,
               ull *__restrict__ received, QueueGeom g) {
   // dynamic LDS: [0..1] control words, [2..] the sort window (the
@@ -254,11 +277,19 @@ __global__ void __launch_bounds__(256)
   if (dyn)
     evict_base = dyn[1];
   const u32 a = agents[b];
-  const ull r = inbox_rpos[a];
+  ull r = inbox_rpos[a];
   const ull w = inbox_wpos[a];
   const u32 nc = carry_n[a];
   const u32 room = g.recv_window - nc;
-  const ull avail = w - r;
+  ull avail = w - r;
+  if (avail > (ull)g.inbox_capacity) {
+    // ring overfilled since the last poll: the oldest (w - r - cap)
+    // entries were overwritten (counted in `dropped` at append time) —
+    // skip to the retained window instead of re-reading live slots as
+    // stale duplicates
+    r = w - (ull)g.inbox_capacity;
+    avail = (ull)g.inbox_capacity;
+  }
   const u32 fresh = (u32)(avail < (ull)room ? avail : (ull)room);
   const u32 total = nc + fresh;
 
@@ -276,9 +307,16 @@ __global__ void __launch_bounds__(256)
         bool vis = true;
         if ((h.vis_mode == VIS_BITMAP || h.vis_mode == VIS_GROUP) &&
             h.bitmap != NO_BITMAP) {
-          const u64 wbits =
-              bitmaps[(u64)h.bitmap * g.bitmap_words + (a >> 6)];
-          vis = (wbits >> (a & 63)) & 1ull;
+          if (bitmap_epochs[h.bitmap] != h.bitmap_epoch) {
+            // the pool slot was recycled since this message was sent:
+            // the original visibility set is gone — hide the message
+            // (exact semantics: never deliver against the wrong bitmap)
+            vis = false;
+          } else {
+            const u64 wbits =
+                bitmaps[(u64)h.bitmap * g.bitmap_words + (a >> 6)];
+            vis = (wbits >> (a & 63)) & 1ull;
+          }
         }
         if (vis)
           key = priority_mode ? prio_key(h.priority, seq) : seq;
@@ -473,6 +511,54 @@ __device__ __forceinline__ u8 fold_c(u8 c, int fold) {
   return (fold && c >= 'A' && c <= 'Z') ? (u8)(c | 0x20) : c;
 }
 
+// packed first-byte test: 0x80 in every byte of the result that equals
+// the splat byte (the classic haszero trick on w ^ splat)
+__device__ __forceinline__ u32 byte_eq_mask(u32 w, u32 splat) {
+  const u32 x = w ^ splat;
+  return (x - 0x01010101u) & ~x & 0x80808080u;
+}
+
+// Candidate verification (rare path): per-byte hitmask in one 16-B
+// chunk, then exact needle compare inside the content window. The
+// chunk's trailing bytes are in L1/L2 — the wave just streamed them.
+__device__ __forceinline__ void verify_chunk(
+    u64 seq, u32 slot, u32 sub, const u8 *__restrict__ needle, int nlen,
+    int fold, const Rec *__restrict__ hdr, const u32 *__restrict__ status,
+    const u8 *__restrict__ payload, u64 *__restrict__ out,
+    u32 *__restrict__ out_count, u32 cap, QueueGeom g) {
+  if (status[slot] == ST_DELETED)
+    return;
+  const u8 n0 = fold_c(needle[0], fold);
+  const Rec h = hdr[slot];
+  const int nstart = (int)h.content_len - nlen + 1;
+  if (nstart <= 0)
+    return;
+  const u8 *text = payload + (u64)slot * g.slot_bytes;
+  const u8 *b = text + ((u64)sub << 4);
+  const int base = (int)(sub << 4);
+  bool found = false;
+  for (int j = 0; j < 16 && !found; ++j) {
+    if (fold_c(b[j], fold) != n0)
+      continue;
+    const int p = base + j;
+    if (p >= nstart)
+      continue;
+    bool m = true;
+    for (int q = 1; q < nlen; ++q) {
+      if (fold_c(text[p + q], fold) != fold_c(needle[q], fold)) {
+        m = false;
+        break;
+      }
+    }
+    found = m;
+  }
+  if (found) {
+    const u32 i = atomicAdd(out_count, 1u);
+    if (i < cap)
+      out[i] = seq;
+  }
+}
+
 __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
                          int nlen, int fold, const Rec *__restrict__ hdr,
                          const u32 *__restrict__ status,
@@ -480,70 +566,91 @@ __global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
                          u64 *__restrict__ out, u32 *__restrict__ out_count,
                          u32 cap, QueueGeom g) {
   // Linear-streaming scan: threads stride 16-B chunks of the slot
-  // REGION in seq order — one coalesced uint4 load per chunk, no
-  // per-message header reads on the hot path. Headers/status load only
-  // when a chunk contains the needle's first byte (~6% of random
-  // chunks for 1-in-256 bytes), so the scan runs at HBM streaming rate
-  // instead of header-latency rate. Slot padding bytes are scanned too
-  // and rejected by the content_len bound at verify time.
-  const u8 n0 = fold_c(needle[0], fold);
+  // REGION in seq order — coalesced uint4 loads, no per-message header
+  // reads on the hot path. Headers/status load only for chunks that
+  // contain the needle's first byte. Two levers put the scan at HBM
+  // streaming rate rather than ALU/latency rate:
+  //  - NO division in the loop: (seq, slot, sub) all step
+  //    incrementally (one div at entry; the old per-chunk 64-bit
+  //    `seq % num_slots` serialized every iteration);
+  //  - 4-deep software pipeline: four independent uint4 loads in
+  //    flight per thread per iteration (one dependent load per
+  //    iteration leaves HBM idle between round-trips);
+  //  - packed-u32 first-byte filter: 4 haszero tests per chunk instead
+  //    of 16 byte compares (2x more with ASCII case folding).
+  // Slot padding bytes are scanned too and rejected by the content_len
+  // bound at verify time.
+  const u8 nf = fold_c(needle[0], fold);
+  const u32 s1 = (u32)nf * 0x01010101u;
+  const bool two = fold && nf >= 'a' && nf <= 'z';
+  const u32 s2 = two ? s1 - 0x20202020u : s1; // upper-case splat
   const u32 cps = g.slot_bytes >> 4; // chunks per slot
   const u64 nchunks = (hi - lo) * cps;
   const u64 stride = (u64)gridDim.x * blockDim.x;
   const u64 ci0 = (u64)blockIdx.x * blockDim.x + threadIdx.x;
   if (ci0 >= nchunks)
     return;
-  // incremental (seq, sub) stepping: one division at entry, none in the
-  // loop (a per-chunk 64-bit div/mod would dominate the scan's VALU)
   u64 seq = lo + ci0 / cps;
   u32 sub = (u32)(ci0 % cps);
+  u32 slot = (u32)(seq % g.num_slots);
   const u64 dseq = stride / cps;
   const u32 dsub = (u32)(stride % cps);
-#pragma unroll 2
-  for (u64 ci = ci0; ci < nchunks; ci += stride) {
-    const u32 slot = (u32)(seq % g.num_slots);
-    const uint4 v = reinterpret_cast<const uint4 *>(
-        payload + (u64)slot * g.slot_bytes)[sub];
-    const u8 *b = reinterpret_cast<const u8 *>(&v);
-    u32 hitmask = 0;
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-      hitmask |= (fold_c(b[j], fold) == n0) ? (1u << j) : 0u;
-    bool found = false;
-    if (hitmask != 0 && status[slot] != ST_DELETED) {
-      // candidate path (rare): check the message and verify positions;
-      // the trailing bytes come from L1/L2 (the wave just streamed them)
-      const Rec h = hdr[slot];
-      const int nstart = (int)h.content_len - nlen + 1;
-      const u8 *text = payload + (u64)slot * g.slot_bytes;
-      const int base = (int)(sub << 4);
-      while (hitmask && !found && nstart > 0) {
-        const int j = __builtin_ctz(hitmask);
-        hitmask &= hitmask - 1;
-        const int p = base + j;
-        if (p >= nstart)
-          continue;
-        bool m = true;
-        for (int q = 1; q < nlen; ++q) {
-          if (fold_c(text[p + q], fold) != fold_c(needle[q], fold)) {
-            m = false;
-            break;
-          }
-        }
-        found = m;
-      }
-      if (found) {
-        const u32 i = atomicAdd(out_count, 1u);
-        if (i < cap)
-          out[i] = seq;
-      }
-    }
-    seq += dseq;
+  const u32 dslot = (u32)(dseq % (u64)g.num_slots);
+
+  // one incremental step of the (seq, slot, sub) walker
+  auto step = [&]() {
     sub += dsub;
+    u32 carry = 0;
     if (sub >= cps) {
       sub -= cps;
-      ++seq;
+      carry = 1;
     }
+    seq += dseq + carry;
+    slot += dslot + carry;
+    if (slot >= g.num_slots)
+      slot -= g.num_slots;
+  };
+  auto test16 = [&](const uint4 &v) -> bool {
+    u32 m = byte_eq_mask(v.x, s1) | byte_eq_mask(v.y, s1) |
+            byte_eq_mask(v.z, s1) | byte_eq_mask(v.w, s1);
+    if (two)
+      m |= byte_eq_mask(v.x, s2) | byte_eq_mask(v.y, s2) |
+           byte_eq_mask(v.z, s2) | byte_eq_mask(v.w, s2);
+    return m != 0;
+  };
+
+  constexpr int P = 4; // pipeline depth
+  u64 ci = ci0;
+  for (; ci + (u64)(P - 1) * stride < nchunks; ci += (u64)P * stride) {
+    u64 seq_p[P];
+    u32 slot_p[P], sub_p[P];
+    const u8 *addr_p[P];
+#pragma unroll
+    for (int p = 0; p < P; ++p) {
+      seq_p[p] = seq;
+      slot_p[p] = slot;
+      sub_p[p] = sub;
+      addr_p[p] =
+          payload + (u64)slot * g.slot_bytes + ((u64)sub << 4);
+      step();
+    }
+    uint4 v[P];
+#pragma unroll
+    for (int p = 0; p < P; ++p) // independent loads, all in flight
+      v[p] = *reinterpret_cast<const uint4 *>(addr_p[p]);
+#pragma unroll
+    for (int p = 0; p < P; ++p)
+      if (test16(v[p]))
+        verify_chunk(seq_p[p], slot_p[p], sub_p[p], needle, nlen, fold,
+                     hdr, status, payload, out, out_count, cap, g);
+  }
+  for (; ci < nchunks; ci += stride) {
+    const uint4 v = *reinterpret_cast<const uint4 *>(
+        payload + (u64)slot * g.slot_bytes + ((u64)sub << 4));
+    if (test16(v))
+      verify_chunk(seq, slot, sub, needle, nlen, fold, hdr, status,
+                   payload, out, out_count, cap, g);
+    step();
   }
 }
 
@@ -713,6 +820,7 @@ public:
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&h2d_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&lb_stream_, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&ev_, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&up_ev_[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&up_ev_[1], hipEventDisableTiming));
@@ -733,6 +841,11 @@ public:
     HIP_CHECK(hipMalloc(&d_active_, (size_t)max_agents * sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_bitmaps_,
                         (size_t)num_bitmaps * g_.bitmap_words * sizeof(u64)));
+    HIP_CHECK(hipMalloc(&d_bitmap_epochs_, (size_t)num_bitmaps * sizeof(u32)));
+    HIP_CHECK(hipMemset(d_bitmap_epochs_, 0xFF,
+                        (size_t)num_bitmaps * sizeof(u32))); // all EPOCH_NONE
+    HIP_CHECK(hipMalloc(&d_dropped_, sizeof(ull)));
+    HIP_CHECK(hipMemset(d_dropped_, 0, sizeof(ull)));
     HIP_CHECK(hipMalloc(&d_by_type_, N_TYPES * sizeof(ull)));
     HIP_CHECK(hipMalloc(&d_by_status_, N_STATUS * sizeof(ull)));
     HIP_CHECK(hipMalloc(&d_sent_, (size_t)max_agents * sizeof(ull)));
@@ -741,6 +854,7 @@ public:
     HIP_CHECK(hipMalloc(&d_bcast_count_, sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_backend_loads_, (size_t)num_backends * sizeof(ull)));
     HIP_CHECK(hipMalloc(&d_choices_, (size_t)staging_batch * sizeof(u32)));
+    HIP_CHECK(hipMalloc(&d_lb_out_, (size_t)staging_batch * sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_match_, (size_t)staging_batch * sizeof(u64)));
     HIP_CHECK(hipMalloc(&d_match_count_, sizeof(u32)));
     HIP_CHECK(hipMalloc(&d_needle_, 256));
@@ -804,9 +918,10 @@ public:
     for (void *p :
          std::vector<void *>{d_hdr_, d_status_, d_payload_, d_inbox_, d_wpos_,
                              d_rpos_, d_carry_, d_carry_n_, d_active_,
-                             d_bitmaps_, d_by_type_, d_by_status_, d_sent_,
+                             d_bitmaps_, d_bitmap_epochs_, d_dropped_,
+                             d_by_type_, d_by_status_, d_sent_,
                              d_received_, d_bcast_, d_bcast_count_,
-                             d_backend_loads_, d_choices_, d_match_,
+                             d_backend_loads_, d_choices_, d_lb_out_, d_match_,
                              d_match_count_, d_needle_, d_agents_, d_unread_,
                              d_dyn_, d_tail_,
                              d_out_seqs_, d_out_counts_, d_stage_recs_[0],
@@ -833,6 +948,7 @@ public:
     (void)hipStreamDestroy(stream_);
     (void)hipStreamDestroy(copy_stream_);
     (void)hipStreamDestroy(h2d_stream_);
+    (void)hipStreamDestroy(lb_stream_);
   }
 
   // ---- registry ----
@@ -1060,21 +1176,23 @@ public:
                          d_tail_, n, (u64)g_.num_slots);
       hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
                          stream_, d_stage_recs_[s], n, 0, d_dyn_, d_hdr_,
-                         d_status_, d_inbox_, d_wpos_, d_by_type_,
-                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_, g_);
+                         d_status_, d_inbox_, d_wpos_, d_rpos_, d_by_type_,
+                         d_by_status_, d_sent_, d_bcast_, d_bcast_count_,
+                         d_dropped_, g_);
       hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
                          stream_, d_stage_recs_[s], d_stage_pay_[s], n, 0,
                          d_dyn_, d_payload_, g_);
       hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256),
                          dim3(256), 0, stream_, d_bcast_, d_bcast_count_,
-                         d_active_, d_hdr_, d_bitmaps_, d_inbox_, d_wpos_,
-                         g_);
+                         d_active_, d_hdr_, d_bitmaps_, d_bitmap_epochs_,
+                         d_inbox_, d_wpos_, d_rpos_, d_dropped_, g_);
       hipLaunchKernelGGL(k_receive, dim3(na), dim3(256),
                          (g_.recv_window + 2) * sizeof(u64), stream_,
                          d_agents_, na, K, priority ? 1 : 0, 0, d_dyn_,
                          d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
-                         d_carry_, d_carry_n_, d_bitmaps_, d_out_seqs_,
-                         d_out_counts_, d_by_status_, d_received_, g_);
+                         d_carry_, d_carry_n_, d_bitmaps_, d_bitmap_epochs_,
+                         d_out_seqs_, d_out_counts_, d_by_status_,
+                         d_received_, g_);
       HIP_CHECK(hipMemcpyAsync(h_out_counts_, d_out_counts_, na * sizeof(u32),
                                hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipMemcpyAsync(h_out_seqs_, d_out_seqs_,
@@ -1174,14 +1292,29 @@ public:
     return base;
   }
 
-  u32 alloc_bitmap(py::buffer words) {
+  // Returns the allocation HANDLE (monotonic counter). The pool slot is
+  // handle % num_bitmaps and the slot's epoch word is set to the handle,
+  // so dequeue-time readers can detect recycling exactly (a message
+  // whose slot was reused is hidden, never filtered against the wrong
+  // bits). Callers put the handle in Rec.bitmap; the engine splits it
+  // into (slot, epoch) before staging.
+  u64 alloc_bitmap(py::buffer words) {
     py::buffer_info wi = words.request();
     if ((size_t)wi.size * wi.itemsize != g_.bitmap_words * sizeof(u64))
       throw std::invalid_argument("bitmap must be max_agents/64 u64 words");
-    const u32 idx = bitmap_next_++ % g_.num_bitmaps;
+    const u32 handle = bitmap_next_++;
+    const u32 idx = handle % g_.num_bitmaps;
+    // invalidate -> write bits -> publish epoch: an in-flight dequeue
+    // never pairs the NEW bits with the OLD epoch (it sees EPOCH_NONE
+    // and hides the message instead)
+    const u32 none = EPOCH_NONE;
+    HIP_CHECK(hipMemcpy(d_bitmap_epochs_ + idx, &none, sizeof(u32),
+                        hipMemcpyHostToDevice));
     HIP_CHECK(hipMemcpy(d_bitmaps_ + (size_t)idx * g_.bitmap_words, wi.ptr,
                         g_.bitmap_words * sizeof(u64), hipMemcpyHostToDevice));
-    return idx;
+    HIP_CHECK(hipMemcpy(d_bitmap_epochs_ + idx, &handle, sizeof(u32),
+                        hipMemcpyHostToDevice));
+    return handle;
   }
 
   // ---- receive plane ----
@@ -1207,8 +1340,9 @@ public:
                          na, max_per_agent, priority ? 1 : 0, evict_base_,
                          (const u64 *)nullptr,
                          d_hdr_, d_status_, d_inbox_, d_wpos_, d_rpos_,
-                         d_carry_, d_carry_n_, d_bitmaps_, d_out_seqs_,
-                         d_out_counts_, d_by_status_, d_received_, g_);
+                         d_carry_, d_carry_n_, d_bitmaps_, d_bitmap_epochs_,
+                         d_out_seqs_, d_out_counts_, d_by_status_,
+                         d_received_, g_);
       HIP_CHECK(hipMemcpyAsync(h_out_counts_, d_out_counts_, na * sizeof(u32),
                                hipMemcpyDeviceToHost, stream_));
       if (return_seqs)
@@ -1549,20 +1683,27 @@ public:
                         g_.max_agents * sizeof(ull), hipMemcpyDeviceToHost));
     HIP_CHECK(hipMemcpy(received.mutable_data(), d_received_,
                         g_.max_agents * sizeof(ull), hipMemcpyDeviceToHost));
+    ull dropped = 0;
+    HIP_CHECK(hipMemcpy(&dropped, d_dropped_, sizeof(ull),
+                        hipMemcpyDeviceToHost));
     py::dict d;
     d["by_type"] = by_type;
     d["by_status"] = by_status;
     d["sent"] = sent;
     d["received"] = received;
+    d["dropped"] = (u64)dropped; // inbox-ring overwrites of unread entries
     return d;
   }
 
   // ---- load balancer ----
 
   void backend_add_load(int idx, long long delta) {
-    hipLaunchKernelGGL(k_add_load, dim3(1), dim3(64), 0, stream_, idx, delta,
-                       d_backend_loads_);
-    HIP_CHECK(hipStreamSynchronize(stream_));
+    // balancer state is only touched by lb kernels — running them on
+    // their own stream keeps dispatch from serializing against the
+    // delivery tick under combined load
+    hipLaunchKernelGGL(k_add_load, dim3(1), dim3(64), 0, lb_stream_, idx,
+                       delta, d_backend_loads_);
+    HIP_CHECK(hipStreamSynchronize(lb_stream_));
   }
 
   py::array_t<i64> backend_loads() {
@@ -1582,12 +1723,12 @@ public:
     py::array_t<u32> out(requests);
     {
       py::gil_scoped_release nogil;
-      hipLaunchKernelGGL(k_lb_batch, dim3(1), dim3(64), 0, stream_, requests,
-                         n_backends, d_backend_loads_, d_choices_);
-      HIP_CHECK(hipMemcpyAsync(h_choices_, d_choices_,
+      hipLaunchKernelGGL(k_lb_batch, dim3(1), dim3(64), 0, lb_stream_,
+                         requests, n_backends, d_backend_loads_, d_lb_out_);
+      HIP_CHECK(hipMemcpyAsync(h_choices_, d_lb_out_,
                                requests * sizeof(u32), hipMemcpyDeviceToHost,
-                               stream_));
-      HIP_CHECK(hipStreamSynchronize(stream_));
+                               lb_stream_));
+      HIP_CHECK(hipStreamSynchronize(lb_stream_));
     }
     std::memcpy(out.mutable_data(), h_choices_, requests * sizeof(u32));
     return out;
@@ -1607,14 +1748,16 @@ private:
   void launch_enqueue(const Rec *recs, const u8 *pay, int n, u64 base) {
     hipLaunchKernelGGL(k_enqueue_meta, dim3((n + 255) / 256), dim3(256), 0,
                        stream_, recs, n, base, (const u64 *)nullptr, d_hdr_,
-                       d_status_, d_inbox_, d_wpos_, d_by_type_, d_by_status_,
-                       d_sent_, d_bcast_, d_bcast_count_, g_);
+                       d_status_, d_inbox_, d_wpos_, d_rpos_, d_by_type_,
+                       d_by_status_, d_sent_, d_bcast_, d_bcast_count_,
+                       d_dropped_, g_);
     hipLaunchKernelGGL(k_enqueue_payload, dim3((n + 3) / 4), dim3(256), 0,
                        stream_, recs, pay, n, base, (const u64 *)nullptr,
                        d_payload_, g_);
     hipLaunchKernelGGL(k_fanout, dim3((g_.max_agents + 255) / 256), dim3(256),
                        0, stream_, d_bcast_, d_bcast_count_, d_active_,
-                       d_hdr_, d_bitmaps_, d_inbox_, d_wpos_, g_);
+                       d_hdr_, d_bitmaps_, d_bitmap_epochs_, d_inbox_,
+                       d_wpos_, d_rpos_, d_dropped_, g_);
   }
 
   void check_agent(u32 idx) const {
@@ -1653,7 +1796,7 @@ private:
   size_t out_pool_ = 0;
   size_t stage_pay_bytes_ = 0;
 
-  hipStream_t stream_{}, copy_stream_{}, h2d_stream_{};
+  hipStream_t stream_{}, copy_stream_{}, h2d_stream_{}, lb_stream_{};
   hipEvent_t ev_{};
   hipEvent_t up_ev_[2] = {};
   hipEvent_t d2h_ev_[2] = {};
@@ -1675,6 +1818,8 @@ private:
   u32 *d_carry_n_{};
   u32 *d_active_{};
   u64 *d_bitmaps_{};
+  u32 *d_bitmap_epochs_{};
+  ull *d_dropped_{};
   ull *d_by_type_{};
   ull *d_by_status_{};
   ull *d_sent_{};
@@ -1683,6 +1828,7 @@ private:
   u32 *d_bcast_count_{};
   ull *d_backend_loads_{};
   u32 *d_choices_{};
+  u32 *d_lb_out_{};
   u64 *d_match_{};
   u32 *d_match_count_{};
   u8 *d_needle_{};
@@ -1708,6 +1854,308 @@ private:
   u32 *h_fetch_status_{};
   u8 *h_fetch_pay_{};
   u32 *h_choices_{};
+};
+
+// ---------------------------------------------------------------------------
+// DoorbellQueue — persistent-kernel express lane for single-message
+// latency (VERDICT round-1 item 5; the batched tick above is the
+// throughput plane). A resident one-wavefront kernel polls a doorbell in
+// pinned host memory; the host writes a message + rings, the kernel
+// copies it into the receiver's pinned delivery ring and bumps a
+// completion word the host spins on. No stream round-trips, no kernel
+// launches on the message path: send->receive is two PCIe hops + one
+// LDS-free wavefront copy.
+//
+// Replaces the reference's latency floor of linger.ms=10 + 1 s consumer
+// polls ("swarmdb/ main.py":197, 557-558) for the p2p single-message
+// regime (BASELINE config 2).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+// control block indices (u64 words in pinned host memory)
+constexpr int DB_HEAD = 0;     // host writes: total submitted
+constexpr int DB_STOP = 1;     // host writes 1 to stop the kernel
+constexpr int DB_CONSUMED = 2; // device writes: total consumed
+constexpr int DB_EXITED = 3;   // device writes 1 on exit
+constexpr int DB_NWORDS = 8;
+
+struct DoorGeom {
+  u32 sub_cap;    // submit ring entries
+  u32 n_agents;   // express agent slots
+  u32 ring_cap;   // delivery ring entries per agent
+  u32 slot_bytes; // payload bytes per entry (16-B multiple)
+};
+
+} // namespace
+
+// One wavefront; lane 0 does the control-word traffic, all 64 lanes
+// cooperate on payload copies. Self-terminates after max_cycles of
+// 100 MHz s_memrealtime ticks so a wedged host can never leave the GPU
+// spinning forever.
+__global__ void k_doorbell(volatile ull *ctrl, const Rec *__restrict__ sub_recs,
+                           const u8 *__restrict__ sub_pay,
+                           Rec *__restrict__ del_recs, u8 *__restrict__ del_pay,
+                           volatile ull *del_count, ull *__restrict__ del_next,
+                           DoorGeom dg, unsigned long long max_cycles) {
+  const int lane = threadIdx.x & 63;
+  const unsigned long long t0 = __builtin_amdgcn_s_memrealtime();
+  u64 consumed = ctrl[DB_CONSUMED];
+  for (;;) {
+    u64 head, stop;
+    if (lane == 0) {
+      head = ctrl[DB_HEAD];
+      stop = ctrl[DB_STOP];
+    }
+    head = __shfl(head, 0, 64);
+    stop = __shfl(stop, 0, 64);
+    if (stop)
+      break;
+    if (head == consumed) {
+      if (__builtin_amdgcn_s_memrealtime() - t0 > max_cycles)
+        break;
+      __builtin_amdgcn_s_sleep(32);
+      continue;
+    }
+    while (consumed < head) {
+      const u32 i = (u32)(consumed % dg.sub_cap);
+      // header: lane 0 reads + broadcasts the routing fields
+      u64 recv_len;
+      if (lane == 0) {
+        const Rec &r = sub_recs[i];
+        recv_len = ((u64)r.receiver << 32) | r.payload_len;
+      }
+      recv_len = __shfl(recv_len, 0, 64);
+      const u32 recv = (u32)(recv_len >> 32);
+      const u32 plen = (u32)recv_len;
+      if (recv < dg.n_agents) {
+        const u64 pos = del_next[recv];
+        const u64 slot = (u64)recv * dg.ring_cap + (pos % dg.ring_cap);
+        // payload copy: 64 lanes x 16 B per round
+        const uint4 *src = reinterpret_cast<const uint4 *>(
+            sub_pay + (u64)i * dg.slot_bytes);
+        uint4 *dst =
+            reinterpret_cast<uint4 *>(del_pay + slot * dg.slot_bytes);
+        const u32 nchunk = (plen + 15u) >> 4;
+        for (u32 c = lane; c < nchunk; c += 64)
+          dst[c] = src[c];
+        if (lane == 0) {
+          del_recs[slot] = sub_recs[i];
+          del_next[recv] = pos + 1;
+        }
+        // payload+header visible on the host BEFORE the counter bump
+        __threadfence_system();
+        if (lane == 0)
+          del_count[recv] = pos + 1;
+      }
+      ++consumed;
+    }
+    if (lane == 0) {
+      ctrl[DB_CONSUMED] = consumed;
+      __threadfence_system();
+    }
+  }
+  if (lane == 0) {
+    ctrl[DB_CONSUMED] = consumed;
+    __threadfence_system();
+    ctrl[DB_EXITED] = 1;
+  }
+}
+
+class DoorbellQueue {
+public:
+  DoorbellQueue(u32 slot_bytes, u32 sub_cap, u32 n_agents, u32 ring_cap,
+                int device)
+      : device_(device) {
+    if (slot_bytes % 16 != 0)
+      throw std::invalid_argument("slot_bytes must be a multiple of 16");
+    dg_.slot_bytes = slot_bytes;
+    dg_.sub_cap = sub_cap;
+    dg_.n_agents = n_agents;
+    dg_.ring_cap = ring_cap;
+    HIP_CHECK(hipSetDevice(device_));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    const unsigned flags = hipHostMallocCoherent | hipHostMallocMapped;
+    HIP_CHECK(hipHostMalloc((void **)&h_ctrl_, DB_NWORDS * sizeof(ull), flags));
+    HIP_CHECK(hipHostMalloc((void **)&h_sub_recs_,
+                            (size_t)sub_cap * sizeof(Rec), flags));
+    HIP_CHECK(hipHostMalloc((void **)&h_sub_pay_,
+                            (size_t)sub_cap * slot_bytes, flags));
+    HIP_CHECK(hipHostMalloc((void **)&h_del_recs_,
+                            (size_t)n_agents * ring_cap * sizeof(Rec), flags));
+    HIP_CHECK(hipHostMalloc((void **)&h_del_pay_,
+                            (size_t)n_agents * ring_cap * slot_bytes, flags));
+    HIP_CHECK(hipHostMalloc((void **)&h_del_count_,
+                            (size_t)n_agents * sizeof(ull), flags));
+    std::memset((void *)h_ctrl_, 0, DB_NWORDS * sizeof(ull));
+    std::memset((void *)h_del_count_, 0, (size_t)n_agents * sizeof(ull));
+    // device-private per-agent delivery cursor (only the kernel touches it)
+    HIP_CHECK(hipMalloc(&d_del_next_, (size_t)n_agents * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_del_next_, 0, (size_t)n_agents * sizeof(ull)));
+    read_pos_.assign(n_agents, 0);
+    HIP_CHECK(hipHostGetDevicePointer((void **)&m_ctrl_, (void *)h_ctrl_, 0));
+    HIP_CHECK(hipHostGetDevicePointer((void **)&m_sub_recs_, h_sub_recs_, 0));
+    HIP_CHECK(hipHostGetDevicePointer((void **)&m_sub_pay_, h_sub_pay_, 0));
+    HIP_CHECK(hipHostGetDevicePointer((void **)&m_del_recs_, h_del_recs_, 0));
+    HIP_CHECK(hipHostGetDevicePointer((void **)&m_del_pay_, h_del_pay_, 0));
+    HIP_CHECK(
+        hipHostGetDevicePointer((void **)&m_del_count_, (void *)h_del_count_, 0));
+  }
+
+  ~DoorbellQueue() { release(); }
+
+  void release() {
+    if (released_)
+      return;
+    released_ = true;
+    if (running_)
+      stop();
+    (void)hipStreamDestroy(stream_);
+    (void)hipFree(d_del_next_);
+    (void)hipHostFree((void *)h_ctrl_);
+    (void)hipHostFree(h_sub_recs_);
+    (void)hipHostFree(h_sub_pay_);
+    (void)hipHostFree(h_del_recs_);
+    (void)hipHostFree(h_del_pay_);
+    (void)hipHostFree((void *)h_del_count_);
+  }
+
+  void start(double max_seconds = 60.0) {
+    if (running_)
+      return;
+    h_ctrl_[DB_STOP] = 0;
+    h_ctrl_[DB_EXITED] = 0;
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    // s_memrealtime ticks at 100 MHz on CDNA
+    const unsigned long long max_cycles =
+        (unsigned long long)(max_seconds * 100.0e6);
+    hipLaunchKernelGGL(k_doorbell, dim3(1), dim3(64), 0, stream_, m_ctrl_,
+                       m_sub_recs_, m_sub_pay_, m_del_recs_, m_del_pay_,
+                       m_del_count_, d_del_next_, dg_, max_cycles);
+    HIP_CHECK(hipGetLastError());
+    running_ = true;
+  }
+
+  void stop() {
+    if (!running_)
+      return;
+    h_ctrl_[DB_STOP] = 1;
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    {
+      py::gil_scoped_release nogil;
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    running_ = false;
+  }
+
+  bool running() const { return running_; }
+  bool exited() const { return h_ctrl_[DB_EXITED] != 0; }
+  u64 consumed() const { return h_ctrl_[DB_CONSUMED]; }
+
+  // Submit one message to the express lane. Spins (bounded) if the ring
+  // is full. Returns the express sequence number.
+  u64 send(u32 receiver, u32 sender, py::buffer payload) {
+    py::buffer_info pi = payload.request();
+    const size_t n = (size_t)pi.size * pi.itemsize;
+    if (n > dg_.slot_bytes)
+      throw std::invalid_argument("payload exceeds the doorbell slot");
+    if (receiver >= dg_.n_agents)
+      throw std::out_of_range("receiver outside the express agent set");
+    const u64 h = sub_head_;
+    // bounded wait for ring space (the kernel is normally far ahead)
+    for (int spins = 0; h - h_ctrl_[DB_CONSUMED] >= dg_.sub_cap; ++spins) {
+      if (spins > 50'000'000)
+        throw std::runtime_error("doorbell ring stalled (kernel dead?)");
+    }
+    const u32 i = (u32)(h % dg_.sub_cap);
+    Rec r{};
+    r.sender = sender;
+    r.receiver = receiver;
+    r.payload_len = (u32)n;
+    r.content_len = (u32)n;
+    std::memcpy(h_sub_pay_ + (size_t)i * dg_.slot_bytes, pi.ptr, n);
+    h_sub_recs_[i] = r;
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    h_ctrl_[DB_HEAD] = h + 1; // the doorbell
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    sub_head_ = h + 1;
+    return h;
+  }
+
+  // Non-blocking poll of an agent's delivery ring. Returns
+  // (sender, payload bytes) or None.
+  py::object try_recv(u32 agent) {
+    if (agent >= dg_.n_agents)
+      throw std::out_of_range("agent outside the express agent set");
+    const u64 have = h_ctrl_exited_safe_count(agent);
+    u64 &rp = read_pos_[agent];
+    if (have == rp)
+      return py::none();
+    if (have - rp > dg_.ring_cap) // overwritten: skip to the oldest intact
+      rp = have - dg_.ring_cap;
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    const u64 slot = (u64)agent * dg_.ring_cap + (rp % dg_.ring_cap);
+    const Rec r = h_del_recs_[slot];
+    py::bytes pay(
+        reinterpret_cast<const char *>(h_del_pay_ + slot * dg_.slot_bytes),
+        r.payload_len);
+    ++rp;
+    return py::make_tuple(r.sender, pay);
+  }
+
+  // Blocking receive with a spin deadline in microseconds. GIL released
+  // while spinning so other threads can send.
+  py::object recv_spin(u32 agent, double timeout_us) {
+    if (agent >= dg_.n_agents)
+      throw std::out_of_range("agent outside the express agent set");
+    u64 &rp = read_pos_[agent];
+    bool got = false;
+    {
+      py::gil_scoped_release nogil;
+      const auto t0 = std::chrono::steady_clock::now();
+      for (;;) {
+        if (h_ctrl_exited_safe_count(agent) != rp) {
+          got = true;
+          break;
+        }
+        const auto el = std::chrono::duration_cast<std::chrono::microseconds>(
+                            std::chrono::steady_clock::now() - t0)
+                            .count();
+        if ((double)el > timeout_us)
+          break;
+      }
+    }
+    if (!got)
+      return py::none();
+    return try_recv(agent);
+  }
+
+private:
+  u64 h_ctrl_exited_safe_count(u32 agent) const {
+    return ((volatile ull *)h_del_count_)[agent];
+  }
+
+  DoorGeom dg_;
+  int device_;
+  bool running_ = false;
+  bool released_ = false;
+  u64 sub_head_ = 0;
+  hipStream_t stream_{};
+  volatile ull *h_ctrl_{};
+  Rec *h_sub_recs_{};
+  u8 *h_sub_pay_{};
+  Rec *h_del_recs_{};
+  u8 *h_del_pay_{};
+  volatile ull *h_del_count_{};
+  ull *d_del_next_{};
+  std::vector<u64> read_pos_;
+  // device-visible mappings of the pinned blocks
+  volatile ull *m_ctrl_{};
+  Rec *m_sub_recs_{};
+  u8 *m_sub_pay_{};
+  Rec *m_del_recs_{};
+  u8 *m_del_pay_{};
+  volatile ull *m_del_count_{};
 };
 
 // ---------------------------------------------------------------------------
@@ -1773,4 +2221,20 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("slot_bytes", &DeviceQueue::slot_bytes)
       .def("recv_window", &DeviceQueue::recv_window)
       .def("release", &DeviceQueue::release);
+
+  py::class_<DoorbellQueue>(m, "DoorbellQueue")
+      .def(py::init<u32, u32, u32, u32, int>(), py::arg("slot_bytes") = 1024,
+           py::arg("sub_cap") = 1024, py::arg("n_agents") = 64,
+           py::arg("ring_cap") = 256, py::arg("device") = 0)
+      .def("start", &DoorbellQueue::start, py::arg("max_seconds") = 60.0)
+      .def("stop", &DoorbellQueue::stop)
+      .def("running", &DoorbellQueue::running)
+      .def("exited", &DoorbellQueue::exited)
+      .def("consumed", &DoorbellQueue::consumed)
+      .def("send", &DoorbellQueue::send, py::arg("receiver"),
+           py::arg("sender"), py::arg("payload"))
+      .def("try_recv", &DoorbellQueue::try_recv)
+      .def("recv_spin", &DoorbellQueue::recv_spin, py::arg("agent"),
+           py::arg("timeout_us") = 1e6)
+      .def("release", &DoorbellQueue::release);
 }

@@ -42,6 +42,9 @@ class QueueConfig:
     #                                      (pow2; small windows keep many
     #                                      dequeue workgroups resident)
     staging_batch: int = 16384           # max messages per enqueue batch
+    num_bitmaps: int = 4096              # visibility-bitmap pool depth
+    #                                      (epoch-tagged ring: recycled
+    #                                      entries are detected exactly)
     num_backends: int = 64               # LLM backend table capacity
     use_gpu: Optional[bool] = None       # None = auto-detect
 
@@ -78,6 +81,7 @@ class QueueConfig:
             num_slots=int(env.get("SWARMQ_NUM_SLOTS", str(1 << 20))),
             inbox_capacity=int(env.get("SWARMQ_INBOX_CAPACITY", str(1 << 16))),
             staging_batch=int(env.get("SWARMQ_STAGING_BATCH", "16384")),
+            num_bitmaps=int(env.get("SWARMQ_NUM_BITMAPS", "4096")),
             device_index=int(env.get("SWARMQ_DEVICE", "0")),
         )
         kw.update(overrides)

@@ -102,6 +102,13 @@ class GpuDirectRouter:
         out_recs["payload_off"] = (dst_off - rank_base[inst_rank]).astype(
             np.uint64
         )
+        # the receive side ingests straight from device memory
+        # (enqueue_from_ptrs bypasses the engine's staging transform), so
+        # split bitmap handles -> (pool slot, epoch) here; the pool is
+        # allocated in lockstep on every rank, so the mapping is global
+        split = getattr(self.engine, "split_bitmap_handles", None)
+        if split is not None:
+            out_recs = split(out_recs)
 
         # exchange section sizes
         sz = torch.tensor(
@@ -239,18 +246,19 @@ class CrossGpuRouter:
                 pay_bytes = dst.tobytes()
             out_blobs.append(sub.tobytes() + pay_bytes)
 
-        sizes = torch.tensor(
-            [len(b) for b in out_blobs], dtype=torch.int64, device=self.device
-        )
-        nmsgs = torch.tensor(
-            [int(((dest == d) | bmask).sum()) for d in range(W)],
+        # one fused header exchange: (blob bytes, msg count) per rank
+        hdr = torch.tensor(
+            [
+                [len(out_blobs[d]), int(((dest == d) | bmask).sum())]
+                for d in range(W)
+            ],
             dtype=torch.int64,
             device=self.device,
-        )
-        in_sizes = torch.empty_like(sizes)
-        in_nmsgs = torch.empty_like(nmsgs)
-        dist.all_to_all_single(in_sizes, sizes, group=self.group)
-        dist.all_to_all_single(in_nmsgs, nmsgs, group=self.group)
+        ).reshape(-1)
+        in_hdr = torch.empty_like(hdr)
+        dist.all_to_all_single(in_hdr, hdr, group=self.group)
+        in_pairs = in_hdr.reshape(W, 2).cpu()
+        in_sizes, in_nmsgs = in_pairs[:, 0], in_pairs[:, 1]
 
         blob = b"".join(out_blobs)
         if blob:
@@ -265,7 +273,7 @@ class CrossGpuRouter:
             recv_buf,
             send_buf,
             output_split_sizes=in_sizes.tolist(),
-            input_split_sizes=sizes.tolist(),
+            input_split_sizes=[len(b) for b in out_blobs],
             group=self.group,
         )
         raw = recv_buf.cpu().numpy().tobytes()

@@ -420,6 +420,7 @@ class CpuEngine(Engine):
                 "by_status": self._by_status.copy(),
                 "sent": self._sent.copy(),
                 "received": self._received.copy(),
+                "dropped": 0,  # CPU inboxes are unbounded
             }
 
     def recv_rate_window(self, agent_idx: int, window_s: float) -> int:

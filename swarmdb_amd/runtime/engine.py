@@ -82,7 +82,7 @@ REC_DTYPE = np.dtype(
         "names": [
             "sender", "receiver", "type", "priority", "vis_mode", "flags",
             "token_count", "timestamp", "payload_off", "payload_len", "bitmap",
-            "content_len", "reserved",
+            "content_len", "bitmap_epoch",
         ],
         "formats": [
             np.uint32, np.uint32, np.uint8, np.uint8, np.uint8, np.uint8,
@@ -127,8 +127,13 @@ class Engine(abc.ABC):
 
     @abc.abstractmethod
     def alloc_bitmap(self, bits: np.ndarray) -> int:
-        """Store a visibility bitmap (bool array [max_agents]); returns its
-        index for REC_DTYPE.bitmap."""
+        """Store a visibility bitmap (bool array [max_agents]); returns an
+        allocation HANDLE for REC_DTYPE.bitmap. On the GPU engine the
+        pool is an epoch-tagged ring: the engine splits the handle into
+        (pool slot, epoch) at enqueue time, and dequeue hides any message
+        whose pool slot was recycled (exact visibility — never filtered
+        against the wrong bitmap). The CPU engine's pool never recycles,
+        so handle == index there."""
 
     # --- receive plane ---
 

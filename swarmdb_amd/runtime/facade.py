@@ -1197,6 +1197,9 @@ class SwarmsDB:
             "messages_by_type": by_type,
             "messages_by_status": by_status,
             "messages_by_agent": by_agent,
+            # inbox-ring overwrites of unread entries under slow
+            # consumers (always 0 on the CPU engine)
+            "messages_dropped": int(stats.get("dropped", 0)),
             "last_save_time": self.last_save_time,
         }
 

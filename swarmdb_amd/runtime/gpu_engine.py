@@ -56,11 +56,12 @@ class GpuEngine(Engine):
             slot_bytes=c.slot_bytes,
             max_agents=c.max_agents,
             inbox_capacity=c.inbox_capacity,
-            # visibility bitmaps live in a ring pool: the most recent
-            # 4096 restricted broadcasts keep exact visibility; older
-            # unread restricted broadcasts may see a recycled bitmap
-            # (retention analog of the slot ring)
-            num_bitmaps=4096,
+            # visibility bitmaps live in an epoch-tagged ring pool:
+            # alloc_bitmap returns a handle; split_bitmap_handles maps it
+            # to (pool slot, epoch) and the dequeue kernel HIDES any
+            # message whose slot was recycled — visibility is exact at
+            # every pool depth (a recycled bitmap is never consulted)
+            num_bitmaps=c.num_bitmaps,
             num_backends=c.num_backends,
             staging_batch=c.staging_batch,
             device=c.device_index,
@@ -68,6 +69,7 @@ class GpuEngine(Engine):
         )
         self._staging = c.staging_batch
         self._slot_bytes = c.slot_bytes
+        self._num_bitmaps = c.num_bitmaps
         # host-side receive-event log (processing_rate probe); one entry
         # per poll tick, pruned by age (see receive_many)
         self._recv_events: List[Tuple[float, np.ndarray, np.ndarray]] = []
@@ -110,11 +112,25 @@ class GpuEngine(Engine):
         out["payload_off"] = new_offs.astype(np.uint64)
         return out, buf.tobytes()
 
+    def split_bitmap_handles(self, recs: np.ndarray) -> np.ndarray:
+        """Map raw bitmap HANDLES (what alloc_bitmap returns and callers
+        put in rec['bitmap']) to (pool slot, epoch) for the kernels.
+        Returns a copy when any row needs the split."""
+        bm = recs["bitmap"]
+        mask = (recs["vis_mode"] != 0) & (bm != 0xFFFFFFFF)
+        if not mask.any():
+            return recs
+        out = recs.copy()
+        out["bitmap_epoch"][mask] = bm[mask]
+        out["bitmap"][mask] = bm[mask] % np.uint32(self._num_bitmaps)
+        return out
+
     def enqueue_batch(self, recs: np.ndarray, payloads: bytes) -> np.ndarray:
         n = len(recs)
         if n == 0:
             return np.empty(0, dtype=np.uint64)
-        recs, payloads = self._pack_aligned(np.ascontiguousarray(recs), payloads)
+        recs = self.split_bitmap_handles(np.ascontiguousarray(recs))
+        recs, payloads = self._pack_aligned(recs, payloads)
         pay_view = np.frombuffer(payloads, dtype=np.uint8)
         with self._lock:
             seqs = np.empty(n, dtype=np.uint64)
@@ -401,6 +417,9 @@ class GpuEngine(Engine):
             "by_status": np.asarray(c["by_status"], dtype=np.int64),
             "sent": np.asarray(c["sent"], dtype=np.int64),
             "received": np.asarray(c["received"], dtype=np.int64),
+            # inbox-ring overwrites of unexamined entries (slow-consumer
+            # loss — the CPU engine's unbounded inboxes never drop)
+            "dropped": int(c["dropped"]),
         }
 
     def recv_rate_window(self, agent_idx: int, window_s: float) -> int:

@@ -731,3 +731,124 @@ def test_deliver_outbuf_bytes(gpu_engine):
     src = np.frombuffer(payload, np.uint8)
     for row, p in zip(hdrs, pays):
         assert len(p) == plen
+
+
+def test_bitmap_epoch_recycling_is_exact():
+    """>pool-depth interleaved restricted broadcasts: messages whose
+    bitmap slot was recycled are HIDDEN (never delivered against the
+    wrong bitmap), newer ones deliver exactly (round-1 weak #3)."""
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    pool = 8
+    eng = GpuEngine(small_cfg(num_bitmaps=pool))
+    n_agents = 32
+    for a in range(n_agents):
+        eng.register_agent(a)
+    # 3x pool depth of restricted broadcasts, each visible to exactly
+    # one distinct agent, all unread until the end
+    n_msgs = 3 * pool
+    handles = []
+    for i in range(n_msgs):
+        bits = np.zeros(256, dtype=bool)
+        bits[i % n_agents] = True
+        handles.append(eng.alloc_bitmap(bits))
+    assert handles == list(range(n_msgs))  # monotonic handles
+    recs = np.zeros(n_msgs, dtype=REC_DTYPE)
+    recs["sender"] = 0
+    recs["receiver"] = BROADCAST
+    recs["vis_mode"] = VIS_BITMAP
+    recs["bitmap"] = np.array(handles, dtype=np.uint32)
+    recs["payload_len"] = 0
+    seqs = eng.enqueue_batch(recs, b"")
+    # only the newest `pool` handles still own their pool slot
+    live = set(handles[-pool:])
+    delivered = {}
+    for a in range(n_agents):
+        got = eng.receive(a, 100)
+        for s in got:
+            delivered[int(s)] = a
+    for i, s in enumerate(seqs):
+        if handles[i] in live:
+            assert delivered.get(int(s)) == i % n_agents, (
+                f"live restricted broadcast {i} must reach exactly its "
+                "member"
+            )
+        else:
+            assert int(s) not in delivered, (
+                f"recycled-bitmap message {i} must be hidden, not "
+                "misdelivered"
+            )
+    eng.close()
+
+
+def test_inbox_overflow_counts_drops():
+    """Overfilling one agent's inbox ring must show up in the dropped
+    counter instead of silently losing messages (advisor finding)."""
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    cap = 1 << 6  # 64-entry inbox ring
+    eng = GpuEngine(small_cfg(inbox_capacity=cap))
+    eng.register_agent(0)
+    eng.register_agent(1)
+    n = 3 * cap
+    recs = np.zeros(n, dtype=REC_DTYPE)
+    recs["sender"] = 1
+    recs["receiver"] = 0
+    recs["vis_mode"] = VIS_ALL
+    recs["bitmap"] = NO_BITMAP
+    recs["payload_len"] = 0
+    eng.enqueue_batch(recs, b"")
+    stats = eng.stats_arrays()
+    assert stats["dropped"] == n - cap  # every overwrite counted
+    got = eng.receive(0, 4 * cap)
+    assert len(got) == cap  # the ring retains the newest cap entries
+    # conservation: retained + counted drops == sent
+    assert len(got) + stats["dropped"] == n
+    eng.close()
+
+
+def test_doorbell_express_latency():
+    """Persistent-kernel express lane: single-message send->receive
+    round trips through pinned mailboxes. Asserts
+    correctness plus a loose latency bound; the measured median is
+    printed for the profile record (<50 us target on MI355X)."""
+    from swarmdb_amd import _swarmq
+
+    db = _swarmq.DoorbellQueue(slot_bytes=1024, sub_cap=256, n_agents=8,
+                               ring_cap=64, device=0)
+    db.start(30.0)
+    try:
+        # functional: payload integrity + sender id + ordering
+        db.send(receiver=1, sender=0, payload=b"hello express")
+        got = db.recv_spin(1, timeout_us=2e6)
+        assert got is not None, "kernel never delivered (doorbell dead)"
+        sender, pay = got
+        assert sender == 0 and bytes(pay) == b"hello express"
+        # burst: 100 messages to one agent arrive in order
+        for i in range(100):
+            db.send(receiver=2, sender=3,
+                    payload=f"m{i:03d}".encode())
+        seen = []
+        for _ in range(100):
+            got = db.recv_spin(2, timeout_us=2e6)
+            assert got is not None
+            seen.append(bytes(got[1]))
+        assert seen == [f"m{i:03d}".encode() for i in range(100)]
+        # latency probe: ping-pong style one-way RTT samples
+        payload = b"x" * 128
+        samples = []
+        for _ in range(200):
+            t0 = time.perf_counter()
+            db.send(receiver=4, sender=5, payload=payload)
+            got = db.recv_spin(4, timeout_us=1e6)
+            dt = time.perf_counter() - t0
+            assert got is not None
+            samples.append(dt)
+        p50 = float(np.median(samples) * 1e6)
+        p99 = float(np.percentile(samples, 99) * 1e6)
+        print(f"\ndoorbell send->receive: p50={p50:.1f}us p99={p99:.1f}us")
+        assert p50 < 1000.0, f"express p50 {p50:.0f}us is not express"
+    finally:
+        db.stop()
+        assert db.exited() or not db.running()
+        db.release()
