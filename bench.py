@@ -9,8 +9,14 @@ Each timed step is one full delivery tick through the GPU-resident queue:
   -> broadcast fan-out kernel -> dequeue kernel (visibility filter +
   LDS sort + READ) -> payload gather kernel -> D2H -> host holds bytes.
 
-A message's send->receive latency is bounded by the tick duration, so the
-reported p50 latency is the median timed step duration.
+Latency is measured PER MESSAGE, not per tick: every message header
+carries its submission timestamp (stamped when the batch is handed to
+the queue — for the pipelined path that is at pinned-buffer staging, so
+pipeline wait is included); on sampled steps the bench drains the
+delivery D2H, records the wall-clock receive time, and keeps a random
+sample of delivered seqs. After the timed loop it fetches those headers
+and reports p50/p99 of (receive time - submission timestamp). The median
+step duration is still reported as ms_per_step.
 
 Usage (driver contract):
   python bench.py --gpus N --steps K --warmup W
@@ -90,6 +96,9 @@ def main() -> int:
                          "(each fans out to every agent)")
     ap.add_argument("--no-gather", action="store_true",
                     help="skip payload D2H gather (delivery stays device-side)")
+    ap.add_argument("--sample-every", type=int, default=4,
+                    help="latency-sample every Nth step (drains the "
+                         "delivery D2H on sampled steps)")
     ap.add_argument("--dump-steps", action="store_true",
                     help="print per-step wall times (variance diagnosis)")
     ap.add_argument("--graph", action="store_true",
@@ -192,6 +201,28 @@ def main() -> int:
     sent_total = 0
     recv_total = 0
 
+    # per-message latency sampling: on every `sample_every`-th step the
+    # delivery D2H is drained, the receive wall-time recorded, and a
+    # random subset of delivered seqs kept; deltas against the header
+    # timestamps are computed after the timed loop
+    sample_every = max(1, args.sample_every)
+    lat_samples: list = []  # (t_recv, seqs ndarray)
+
+    def keep_sample(t_recv: float, seqs, counts=None, K=None) -> None:
+        if seqs is None or len(seqs) == 0:
+            return
+        seqs = np.asarray(seqs, dtype=np.uint64)
+        if counts is not None:
+            # dense [na, K] layout: keep only valid entries
+            c64 = np.asarray(counts, dtype=np.int64)
+            mat = seqs.reshape(len(c64), K)
+            seqs = mat[np.arange(K)[None, :] < c64[:, None]]
+            if len(seqs) == 0:
+                return
+        take = min(128, len(seqs))
+        idx = rng.choice(len(seqs), take, replace=False)
+        lat_samples.append((t_recv, seqs[idx].copy()))
+
     def barrier_sync():
         if dist_on:
             import torch.distributed as dist
@@ -203,6 +234,7 @@ def main() -> int:
     def step(i: int) -> int:
         nonlocal sent_total, recv_total, grouter, router
         recs, payload = batches[i % len(batches)]
+        recs["timestamp"] = time.time()  # submission stamp (per message)
         if grouter is not None:
             try:
                 sent_local = grouter.route_and_enqueue(recs, payload)
@@ -241,6 +273,8 @@ def main() -> int:
                 deliver(seqs, args.payload)
             else:
                 engine.fetch(seqs)
+        if i % sample_every == 0 and ndel:
+            keep_sample(time.time(), seqs)
         sent_total += sent_local
         recv_total += ndel
         return ndel
@@ -261,10 +295,13 @@ def main() -> int:
             np.frombuffer(pp, dtype=np.uint8)[:] = np.frombuffer(
                 payload, dtype=np.uint8
             )
-            pinned.append((pr, pp, len(recs), len(payload)))
+            pinned.append((pr, pp, len(recs), len(payload), pr_view))
 
         def _prefetch(slot: int, i: int) -> None:
-            pr, pp, n, nbytes = pinned[i % len(pinned)]
+            pr, pp, n, nbytes, pr_view = pinned[i % len(pinned)]
+            # submission stamp: the moment the batch is handed to the
+            # queue (upload begins) — pipeline wait counts as latency
+            pr_view["timestamp"] = time.time()
             q.prefetch_from(
                 slot,
                 pr.__array_interface__["data"][0],
@@ -293,22 +330,27 @@ def main() -> int:
             nonlocal sent_total, recv_total
             cur = _cur[0]
             n_staged = len(batches[i % len(batches)][0])
+            sampled = i % sample_every == 0
+            seqs = None
+            counts_dense = None
             if use_graph:
                 # upload batch i+1 first so its H2D overlaps this tick's
                 # graph execution
                 _prefetch(1 - cur, i + 1)
-                counts, _ = q.run_tick(cur)
+                counts, seqs = q.run_tick(cur)
+                counts_dense = counts
                 ndel = int(counts.astype(np.int64).sum())
             else:
                 q.enqueue_staged(cur)
                 # upload batch i+1 on the H2D stream: overlaps batch i's
                 # kernels and the delivery D2H (full-duplex PCIe)
                 _prefetch(1 - cur, i + 1)
-                want_seqs = args.bcast_frac > 0
+                want_seqs = args.bcast_frac > 0 or sampled
                 counts, seqs = q.receive_many(
                     local_agents32, recv_K, bool(args.priority),
                     return_seqs=want_seqs,
                 )
+                counts_dense = counts if want_seqs else None
                 ndel = int(counts.astype(np.int64).sum())
             if not args.no_gather and ndel:
                 stride16 = (args.payload + 15) // 16 * 16
@@ -328,6 +370,12 @@ def main() -> int:
                     mat = seqs.reshape(len(local_agents32), recv_K)
                     taken = mat[np.arange(recv_K)[None, :] < c64[:, None]]
                     engine.deliver_payloads(taken, args.payload)
+            if sampled and ndel and seqs is not None:
+                # drain the in-flight delivery D2H so "received" means
+                # bytes are host-visible, then stamp the receive time
+                engine.delivery_sync()
+                keep_sample(time.time(), seqs,
+                            counts=counts_dense, K=recv_K)
             sent_total += n_staged
             recv_total += ndel
             _cur[0] = 1 - cur
@@ -381,7 +429,31 @@ def main() -> int:
 
     msgs_per_s = recv_all / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
-    p50_ms = float(np.median(step_times) * 1000.0)
+
+    # per-message latency: receive wall time minus the submission stamp
+    # carried in each sampled message's header (computed OUTSIDE the
+    # timed region; evicted samples — zeroed headers — are filtered)
+    p50_ms = p99_ms = None
+    n_lat = 0
+    if lat_samples:
+        deltas = []
+        for t_recv, sample_seqs in lat_samples:
+            hdrs, _ = engine.fetch(sample_seqs)
+            ts = np.asarray(hdrs["timestamp"], dtype=np.float64)
+            d = t_recv - ts
+            d = d[(ts > 0) & (d >= 0) & (d < 60.0)]
+            if len(d):
+                deltas.append(d)
+        if deltas:
+            alld = np.concatenate(deltas)
+            n_lat = int(len(alld))
+            p50_ms = float(np.median(alld) * 1000.0)
+            p99_ms = float(np.percentile(alld, 99) * 1000.0)
+    if p50_ms is None:
+        # no samples (e.g. --sample-every > steps): fall back to the
+        # tick-duration bound and say so via n_lat_samples = 0
+        p50_ms = float(np.median(step_times) * 1000.0)
+        p99_ms = float(np.percentile(step_times, 99) * 1000.0)
 
     if rank == 0:
         out = {
@@ -399,6 +471,8 @@ def main() -> int:
             "dtype": "uint8",
             "data": "synthetic",
             "p50_latency_ms": round(p50_ms, 3),
+            "p99_latency_ms": round(p99_ms, 3),
+            "n_latency_samples": n_lat,
             "config": {
                 "model": "gpu-mpmc-ring-queue",
                 "global_batch": args.batch * world,

@@ -51,12 +51,17 @@ class GpuDirectRouter:
     """
 
     def __init__(self, engine, device: torch.device,
-                 group: Optional[object] = None):
+                 group: Optional[object] = None,
+                 force_exchange: bool = False):
         self.engine = engine
         self.device = device
         self.group = group
         self.world = dist.get_world_size(group)
         self.rank = dist.get_rank(group)
+        # force_exchange runs the full pack-kernel -> RCCL all_to_all ->
+        # device-ingest pipeline even at world 1 (self-exchange) so a
+        # 1-GPU box can execute and test the exact multi-GPU code path
+        self.force_exchange = force_exchange
         # keep recv buffers alive while async ingest kernels may still
         # read them (2 ticks deep)
         self._hold: list = []
@@ -64,7 +69,7 @@ class GpuDirectRouter:
     def route_and_enqueue(self, recs: np.ndarray, payloads: bytes) -> int:
         W = self.world
         q = self.engine.q
-        if W == 1:
+        if W == 1 and not self.force_exchange:
             self.engine.enqueue_batch(recs, payloads)
             return len(recs)
 

@@ -1,0 +1,226 @@
+"""Multi-rank GPU tests of the RCCL cross-GPU plane (VERDICT round-1
+item 1): the GpuDirectRouter pipeline — pack kernel -> RCCL
+``all_to_all_single`` over device buffers -> ``enqueue_from_ptrs`` device
+ingest -> dequeue — executed on real hardware.
+
+Two layers:
+
+- world-1 self-exchange (always runs on a 1-GPU box): RCCL initialized,
+  the full pipeline including the collective executes, conservation and
+  payload integrity asserted;
+- world-2 on ONE device (two ranks, both on cuda:0): real inter-rank
+  RCCL traffic. RCCL may refuse two ranks on one device — the test
+  skips with the library's own error in that case (the driver's 8-GPU
+  round-end bench covers the true multi-device run).
+"""
+
+import json
+import os
+import subprocess
+import sys
+import textwrap
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def _mk_batch(rng, n, agents_global, local_agents, plen, n_bcast):
+    from swarmdb_amd.runtime.engine import (
+        BROADCAST,
+        NO_BITMAP,
+        REC_DTYPE,
+        VIS_ALL,
+    )
+
+    recs = np.zeros(n, dtype=REC_DTYPE)
+    recs["sender"] = rng.choice(local_agents, n)
+    recv = rng.integers(0, agents_global, n).astype(np.uint32)
+    recv[:n_bcast] = BROADCAST
+    recs["receiver"] = recv
+    recs["type"] = 0
+    recs["priority"] = 1
+    recs["vis_mode"] = VIS_ALL
+    recs["bitmap"] = NO_BITMAP
+    recs["payload_len"] = plen
+    recs["content_len"] = plen
+    stride = (plen + 15) // 16 * 16
+    recs["payload_off"] = np.arange(n, dtype=np.uint64) * stride
+    return recs, stride
+
+
+def test_rccl_gpu_direct_world1_self_exchange():
+    """Full GPU-direct pipeline at world 1: pack kernel, RCCL
+    all_to_all_single on device tensors, zero-copy device ingest,
+    dequeue. Conservation + payload integrity asserted."""
+    import torch
+    import torch.distributed as dist
+
+    from swarmdb_amd import QueueConfig
+    from swarmdb_amd.parallel.router import GpuDirectRouter
+    from swarmdb_amd.runtime.engine import BROADCAST
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    assert torch.cuda.is_available()
+    store = dist.TCPStore("127.0.0.1", 29531, 1, True)
+    dist.init_process_group("nccl", store=store, rank=0, world_size=1)
+    try:
+        torch.cuda.set_device(0)
+        cfg = QueueConfig(
+            use_gpu=True, max_agents=64, num_slots=1 << 14, slot_bytes=512,
+            inbox_capacity=1 << 12, staging_batch=4096, auto_save=False,
+        )
+        eng = GpuEngine(cfg)
+        agents = np.arange(16, dtype=np.uint32)
+        for a in agents:
+            eng.register_agent(int(a))
+        router = GpuDirectRouter(
+            eng, torch.device("cuda", 0), force_exchange=True
+        )
+        rng = np.random.default_rng(7)
+        n, plen, n_bcast = 256, 64, 8
+        recs, stride = _mk_batch(rng, n, 16, agents, plen, n_bcast)
+        payload = b"".join(
+            (b"w1m%04d" % i).ljust(stride, b".") for i in range(n)
+        )
+        ingested = router.route_and_enqueue(recs, payload)
+        assert ingested == n
+        counts, seqs = eng.receive_many(agents, 2 * n)
+        delivered = int(counts.sum())
+        n_p2p = n - n_bcast
+        assert delivered == n_p2p + n_bcast * len(agents), (
+            delivered, n_p2p, n_bcast)
+        # payload integrity after the device round-trip
+        hdrs, pays = eng.fetch(np.asarray(seqs[:64], dtype=np.uint64))
+        for row, pay in zip(hdrs, pays):
+            assert pay[:3] == b"w1m" and pay.endswith(b"."), pay[:16]
+        eng.close()
+    finally:
+        dist.destroy_process_group()
+
+
+WORKER2 = textwrap.dedent(
+    """
+    import json, os, sys
+    sys.path.insert(0, %r)
+    import numpy as np
+    import torch
+    import torch.distributed as dist
+
+    from swarmdb_amd import QueueConfig
+    from swarmdb_amd.parallel.router import GpuDirectRouter
+    from swarmdb_amd.runtime.engine import BROADCAST, NO_BITMAP, REC_DTYPE, VIS_ALL
+    from swarmdb_amd.runtime.gpu_engine import GpuEngine
+
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    # BOTH ranks on device 0 (single physical GPU)
+    torch.cuda.set_device(0)
+    try:
+        dist.init_process_group(
+            backend="nccl", device_id=torch.device("cuda", 0)
+        )
+        # surface duplicate-GPU refusal at init time, not mid-pipeline
+        t = torch.ones(4, device="cuda")
+        dist.all_reduce(t)
+        torch.cuda.synchronize()
+        assert float(t.sum().item()) == 4.0 * world
+    except Exception as e:
+        print("RCCL_INIT_FAILED: %%s" %% (e,), flush=True)
+        sys.exit(42)
+
+    cfg = QueueConfig(
+        use_gpu=True, max_agents=64, num_slots=1 << 14, slot_bytes=512,
+        inbox_capacity=1 << 12, staging_batch=4096, auto_save=False,
+        world_size=world, rank=rank,
+    )
+    eng = GpuEngine(cfg)
+    agents_global = 16
+    local_agents = np.arange(rank, agents_global, world, dtype=np.uint32)
+    for a in local_agents:
+        eng.register_agent(int(a))
+    router = GpuDirectRouter(eng, torch.device("cuda", 0))
+
+    rng = np.random.default_rng(100 + rank)
+    n, plen, n_bcast = 200, 64, 4
+    stride = (plen + 15) // 16 * 16
+    recs = np.zeros(n, dtype=REC_DTYPE)
+    recs["sender"] = rng.choice(local_agents, n)
+    recv = rng.integers(0, agents_global, n).astype(np.uint32)
+    recv[:n_bcast] = BROADCAST
+    recs["receiver"] = recv
+    recs["vis_mode"] = VIS_ALL
+    recs["bitmap"] = NO_BITMAP
+    recs["payload_len"] = plen
+    recs["content_len"] = plen
+    recs["payload_off"] = np.arange(n, dtype=np.uint64) * stride
+    payload = b"".join(
+        (b"r%%dm%%04d" %% (rank, i)).ljust(stride, b".")
+        for i in range(n)
+    )
+
+    ticks = 3
+    sent_p2p = 0
+    delivered = 0
+    for _ in range(ticks):
+        router.route_and_enqueue(recs, payload)
+        counts, seqs = eng.receive_many(local_agents, 4 * n)
+        delivered += int(counts.sum())
+        sent_p2p += n - n_bcast
+    torch.cuda.synchronize()
+
+    # payload integrity: fetch a sample of delivered messages
+    hdrs, pays = eng.fetch(np.asarray(seqs[:64], dtype=np.uint64))
+    for row, pay in zip(hdrs, pays):
+        assert pay[:1] == b"r" and pay.endswith(b"."), pay[:16]
+        # messages from BOTH ranks must land here (cross-rank traffic)
+    senders = set(int(p[1:2].decode() or 0) for p in pays if p[:1] == b"r")
+
+    t = torch.tensor([delivered, sent_p2p], dtype=torch.int64,
+                     device="cuda")
+    dist.all_reduce(t)
+    total_delivered, total_p2p = int(t[0].item()), int(t[1].item())
+    # conservation: every p2p delivered exactly once; each broadcast
+    # reaches every registered agent node-wide
+    expect = total_p2p + ticks * n_bcast * world * agents_global
+    assert total_delivered == expect, (total_delivered, expect)
+
+    if rank == 0:
+        print(json.dumps({"ok": True, "delivered": total_delivered,
+                          "world": world}))
+    dist.destroy_process_group()
+    """
+) % str(REPO)
+
+
+def test_rccl_gpu_direct_world2_single_device(tmp_path):
+    """Two ranks, one physical GPU, real RCCL collectives between them:
+    the GpuDirectRouter exchange with conservation asserts. Skips with
+    RCCL's own message if the library refuses two ranks on one device."""
+    import torch
+
+    assert torch.cuda.is_available()
+    script = tmp_path / "worker2.py"
+    script.write_text(WORKER2)
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr=127.0.0.1", "--master-port=29532",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=420, env=env,
+        cwd=str(REPO),
+    )
+    out = proc.stdout + proc.stderr
+    if "RCCL_INIT_FAILED" in out:
+        pytest.skip("RCCL refused 2 ranks on one device: "
+                    + out.split("RCCL_INIT_FAILED:")[1][:200])
+    assert proc.returncode == 0, out
+    assert '"ok": true' in out.lower()
