@@ -429,6 +429,20 @@ def config7_checkpoint(seconds: float) -> None:
     save_s = time.perf_counter() - t0
     size_mb = Path(path).stat().st_size / 1e6
 
+    # binary checkpoint plane (device-gather speed, zero per-message
+    # Python) + delta append + reload
+    t0 = time.perf_counter()
+    bpath = db.save_checkpoint()
+    bsave_s = time.perf_counter() - t0
+    bsize_mb = Path(bpath).stat().st_size / 1e6
+    extra = 1 << 14
+    recs, payload = _make_batch(rng, extra, idxs, idxs, 256)
+    db.send_batch(recs, payload)
+    t0 = time.perf_counter()
+    _, n_delta = db.save_checkpoint_delta()
+    bdelta_s = time.perf_counter() - t0
+    assert n_delta == extra
+
     cfg2 = QueueConfig(use_gpu=_gpu_available(), auto_save=False,
                        max_agents=256, num_slots=n_msgs, slot_bytes=512,
                        staging_batch=1 << 14, save_dir=tmp)
@@ -437,6 +451,17 @@ def config7_checkpoint(seconds: float) -> None:
     loaded = db2.load_message_history(path)
     load_s = time.perf_counter() - t0
     assert loaded == n_msgs, (loaded, n_msgs)
+
+    cfg3 = QueueConfig(use_gpu=_gpu_available(), auto_save=False,
+                       max_agents=256, num_slots=2 * n_msgs, slot_bytes=512,
+                       staging_batch=1 << 14, save_dir=tmp)
+    db3 = SwarmsDB(config=cfg3)
+    t0 = time.perf_counter()
+    bloaded = db3.load_checkpoint(bpath)
+    bload_s = time.perf_counter() - t0
+    assert bloaded == n_msgs + extra, (bloaded, n_msgs + extra)
+    db3.config.auto_save = False
+    db3.close()
 
     db.config.auto_save = False
     db2.config.auto_save = False
@@ -452,6 +477,10 @@ def config7_checkpoint(seconds: float) -> None:
         "save_msgs_per_s": round(n_msgs / save_s, 0),
         "load_s": round(load_s, 2),
         "load_msgs_per_s": round(n_msgs / load_s, 0),
+        "binary_ckpt_mb": round(bsize_mb, 1),
+        "binary_save_msgs_per_s": round(n_msgs / bsave_s, 0),
+        "binary_delta_msgs_per_s": round(extra / bdelta_s, 0),
+        "binary_load_msgs_per_s": round((n_msgs + extra) / bload_s, 0),
         "engine": "gpu" if _gpu_available() else "cpu",
     }))
 

@@ -25,6 +25,7 @@
 #include <algorithm>
 #include <atomic>
 #include <chrono>
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -518,28 +519,38 @@ __device__ __forceinline__ u32 byte_eq_mask(u32 w, u32 splat) {
   return (x - 0x01010101u) & ~x & 0x80808080u;
 }
 
-// Candidate verification (rare path): per-byte hitmask in one 16-B
-// chunk, then exact needle compare inside the content window. The
-// chunk's trailing bytes are in L1/L2 — the wave just streamed them.
+// compact the 0x80-bits of a byte_eq_mask result into 4 low bits
+// (movemask): t*0x00204081 lands byte k's bit at position 28+k
+__device__ __forceinline__ u32 movemask4(u32 t) {
+  return (t * 0x00204081u) >> 28;
+}
+
+// Candidate verification (rare path): the chunk is still in REGISTERS
+// (v) — build the per-byte first-char hitmask from it (no memory
+// re-reads) and only the actual hit positions compare the needle tail
+// against global memory (in L1/L2 — the wave just streamed it).
 __device__ __forceinline__ void verify_chunk(
-    u64 seq, u32 slot, u32 sub, const u8 *__restrict__ needle, int nlen,
-    int fold, const Rec *__restrict__ hdr, const u32 *__restrict__ status,
+    u64 seq, u32 slot, u32 sub, const uint4 &v,
+    const u8 *__restrict__ needle, int nlen, int fold,
+    const Rec *__restrict__ hdr, const u32 *__restrict__ status,
     const u8 *__restrict__ payload, u64 *__restrict__ out,
     u32 *__restrict__ out_count, u32 cap, QueueGeom g) {
-  if (status[slot] == ST_DELETED)
-    return;
   const u8 n0 = fold_c(needle[0], fold);
+  const u8 *bytes = reinterpret_cast<const u8 *>(&v);
+  u32 hitmask = 0;
+#pragma unroll
+  for (int j = 0; j < 16; ++j)
+    hitmask |= (fold_c(bytes[j], fold) == n0) ? (1u << j) : 0u;
+  if (hitmask == 0 || status[slot] == ST_DELETED)
+    return;
   const Rec h = hdr[slot];
   const int nstart = (int)h.content_len - nlen + 1;
-  if (nstart <= 0)
-    return;
   const u8 *text = payload + (u64)slot * g.slot_bytes;
-  const u8 *b = text + ((u64)sub << 4);
   const int base = (int)(sub << 4);
   bool found = false;
-  for (int j = 0; j < 16 && !found; ++j) {
-    if (fold_c(b[j], fold) != n0)
-      continue;
+  while (hitmask && !found && nstart > 0) {
+    const int j = __builtin_ctz(hitmask);
+    hitmask &= hitmask - 1;
     const int p = base + j;
     if (p >= nstart)
       continue;
@@ -559,98 +570,104 @@ __device__ __forceinline__ void verify_chunk(
   }
 }
 
-__global__ void k_search(u64 lo, u64 hi, const u8 *__restrict__ needle,
-                         int nlen, int fold, const Rec *__restrict__ hdr,
+// Linear-streaming scan over ONE contiguous slot segment. The slot
+// region of a seq range is always at most TWO contiguous byte ranges
+// (the ring wraps at most once), so the host splits the scan and each
+// launch is a PURE linear walk: addr = seg + 16*ci — the exact access
+// shape that measures 6.2 TB/s on this chip (csrc/bench_membw.hip),
+// where a per-chunk (slot, sub) walker with 64-bit address math topped
+// out at 2.9 TB/s. Headers/status load only for chunks that pass the
+// packed first-byte + second-byte screen (~P(char)^2 of chunks). Slot
+// padding bytes are scanned too and rejected by the content_len bound
+// at verify time.
+template <int P>
+__global__ void k_search(const u8 *__restrict__ seg, u64 nchunks, u64 seq0,
+                         u32 slot0, const u8 *__restrict__ needle, int nlen,
+                         int fold, const Rec *__restrict__ hdr,
                          const u32 *__restrict__ status,
                          const u8 *__restrict__ payload,
                          u64 *__restrict__ out, u32 *__restrict__ out_count,
                          u32 cap, QueueGeom g) {
-  // Linear-streaming scan: threads stride 16-B chunks of the slot
-  // REGION in seq order — coalesced uint4 loads, no per-message header
-  // reads on the hot path. Headers/status load only for chunks that
-  // contain the needle's first byte. Two levers put the scan at HBM
-  // streaming rate rather than ALU/latency rate:
-  //  - NO division in the loop: (seq, slot, sub) all step
-  //    incrementally (one div at entry; the old per-chunk 64-bit
-  //    `seq % num_slots` serialized every iteration);
-  //  - 4-deep software pipeline: four independent uint4 loads in
-  //    flight per thread per iteration (one dependent load per
-  //    iteration leaves HBM idle between round-trips);
-  //  - packed-u32 first-byte filter: 4 haszero tests per chunk instead
-  //    of 16 byte compares (2x more with ASCII case folding).
-  // Slot padding bytes are scanned too and rejected by the content_len
-  // bound at verify time.
   const u8 nf = fold_c(needle[0], fold);
   const u32 s1 = (u32)nf * 0x01010101u;
   const bool two = fold && nf >= 'a' && nf <= 'z';
   const u32 s2 = two ? s1 - 0x20202020u : s1; // upper-case splat
+  // second-byte screen (nlen >= 2): a chunk is only a candidate if it
+  // also contains the needle's SECOND char (or its first-char hit is in
+  // the final byte, whose successor lives in the next chunk)
+  const u8 ng = nlen >= 2 ? fold_c(needle[1], fold) : 0;
+  const u32 t1 = (u32)ng * 0x01010101u;
+  const bool two2 = fold && ng >= 'a' && ng <= 'z';
+  const u32 t2 = two2 ? t1 - 0x20202020u : t1;
   const u32 cps = g.slot_bytes >> 4; // chunks per slot
-  const u64 nchunks = (hi - lo) * cps;
-  const u64 stride = (u64)gridDim.x * blockDim.x;
-  const u64 ci0 = (u64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (ci0 >= nchunks)
-    return;
-  u64 seq = lo + ci0 / cps;
-  u32 sub = (u32)(ci0 % cps);
-  u32 slot = (u32)(seq % g.num_slots);
-  const u64 dseq = stride / cps;
-  const u32 dsub = (u32)(stride % cps);
-  const u32 dslot = (u32)(dseq % (u64)g.num_slots);
 
-  // one incremental step of the (seq, slot, sub) walker
-  auto step = [&]() {
-    sub += dsub;
-    u32 carry = 0;
-    if (sub >= cps) {
-      sub -= cps;
-      carry = 1;
-    }
-    seq += dseq + carry;
-    slot += dslot + carry;
-    if (slot >= g.num_slots)
-      slot -= g.num_slots;
-  };
+  // three-tier screen, costed for wave-level issue (a divergent branch
+  // is paid whenever ANY lane takes it):
+  //   tier 1 (every chunk): does the chunk contain the first char at
+  //     all — 4-8 haszero tests;
+  //   tier 2 (~16% of chunks on random text): EXACT positional check
+  //     that some first-char hit is immediately followed by the second
+  //     char (movemask + shifted AND) — not just "second char appears
+  //     somewhere";
+  //   tier 3 (~needle-density of chunks, ~0.2% random): full verify
+  //     with header/status loads.
+  // The approximate contains-n1 screen left ~2.4% of chunks entering
+  // tier 3, whose divergent dependent loads cost ~1 TB/s (measured,
+  // profiles/r02 search notes).
   auto test16 = [&](const uint4 &v) -> bool {
-    u32 m = byte_eq_mask(v.x, s1) | byte_eq_mask(v.y, s1) |
-            byte_eq_mask(v.z, s1) | byte_eq_mask(v.w, s1);
-    if (two)
-      m |= byte_eq_mask(v.x, s2) | byte_eq_mask(v.y, s2) |
-           byte_eq_mask(v.z, s2) | byte_eq_mask(v.w, s2);
-    return m != 0;
+    u32 e0 = byte_eq_mask(v.x, s1), e1 = byte_eq_mask(v.y, s1),
+        e2 = byte_eq_mask(v.z, s1), e3 = byte_eq_mask(v.w, s1);
+    if (two) {
+      e0 |= byte_eq_mask(v.x, s2);
+      e1 |= byte_eq_mask(v.y, s2);
+      e2 |= byte_eq_mask(v.z, s2);
+      e3 |= byte_eq_mask(v.w, s2);
+    }
+    if ((e0 | e1 | e2 | e3) == 0)
+      return false;
+    if (nlen < 2)
+      return true;
+    u32 f0 = byte_eq_mask(v.x, t1), f1 = byte_eq_mask(v.y, t1),
+        f2 = byte_eq_mask(v.z, t1), f3 = byte_eq_mask(v.w, t1);
+    if (two2) {
+      f0 |= byte_eq_mask(v.x, t2);
+      f1 |= byte_eq_mask(v.y, t2);
+      f2 |= byte_eq_mask(v.z, t2);
+      f3 |= byte_eq_mask(v.w, t2);
+    }
+    const u32 m0 = movemask4(e0) | (movemask4(e1) << 4) |
+                   (movemask4(e2) << 8) | (movemask4(e3) << 12);
+    const u32 m1 = movemask4(f0) | (movemask4(f1) << 4) |
+                   (movemask4(f2) << 8) | (movemask4(f3) << 12);
+    // bit 15 (chunk's last byte) always passes: its successor lives in
+    // the next chunk and is checked by the verify's tail compare
+    return (m0 & ((m1 >> 1) | 0x8000u)) != 0;
+  };
+  auto verify = [&](u64 ci, const uint4 &v) {
+    // rare path: division happens HERE, never in the hot loop
+    const u64 q = ci / cps;
+    const u32 sub = (u32)(ci - q * cps);
+    verify_chunk(seq0 + q, slot0 + (u32)q, sub, v, needle, nlen, fold,
+                 hdr, status, payload, out, out_count, cap, g);
   };
 
-  constexpr int P = 4; // pipeline depth
-  u64 ci = ci0;
+  const u64 stride = (u64)gridDim.x * blockDim.x;
+  const uint4 *src = reinterpret_cast<const uint4 *>(seg);
+  u64 ci = (u64)blockIdx.x * blockDim.x + threadIdx.x;
   for (; ci + (u64)(P - 1) * stride < nchunks; ci += (u64)P * stride) {
-    u64 seq_p[P];
-    u32 slot_p[P], sub_p[P];
-    const u8 *addr_p[P];
-#pragma unroll
-    for (int p = 0; p < P; ++p) {
-      seq_p[p] = seq;
-      slot_p[p] = slot;
-      sub_p[p] = sub;
-      addr_p[p] =
-          payload + (u64)slot * g.slot_bytes + ((u64)sub << 4);
-      step();
-    }
     uint4 v[P];
 #pragma unroll
     for (int p = 0; p < P; ++p) // independent loads, all in flight
-      v[p] = *reinterpret_cast<const uint4 *>(addr_p[p]);
+      v[p] = src[ci + (u64)p * stride];
 #pragma unroll
     for (int p = 0; p < P; ++p)
       if (test16(v[p]))
-        verify_chunk(seq_p[p], slot_p[p], sub_p[p], needle, nlen, fold,
-                     hdr, status, payload, out, out_count, cap, g);
+        verify(ci + (u64)p * stride, v[p]);
   }
   for (; ci < nchunks; ci += stride) {
-    const uint4 v = *reinterpret_cast<const uint4 *>(
-        payload + (u64)slot * g.slot_bytes + ((u64)sub << 4));
+    const uint4 v = src[ci];
     if (test16(v))
-      verify_chunk(seq, slot, sub, needle, nlen, fold, hdr, status,
-                   payload, out, out_count, cap, g);
-    step();
+      verify(ci, v);
   }
 }
 
@@ -1317,6 +1334,23 @@ public:
     return handle;
   }
 
+  // Read back one visibility-bitmap pool slot: (epoch, words bytes).
+  // Checkpoint save uses this to persist the live bitmaps its records
+  // reference (epoch mismatch => the caller's record is already hidden).
+  py::tuple get_bitmap(u32 idx) {
+    if (idx >= g_.num_bitmaps)
+      throw std::out_of_range("bitmap index out of range");
+    u32 epoch = 0;
+    std::vector<u64> words(g_.bitmap_words);
+    HIP_CHECK(hipMemcpy(&epoch, d_bitmap_epochs_ + idx, sizeof(u32),
+                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(words.data(), d_bitmaps_ + (size_t)idx * g_.bitmap_words,
+                        g_.bitmap_words * sizeof(u64), hipMemcpyDeviceToHost));
+    return py::make_tuple(
+        epoch, py::bytes(reinterpret_cast<const char *>(words.data()),
+                         g_.bitmap_words * sizeof(u64)));
+  }
+
   // ---- receive plane ----
 
   // Returns (counts[n_agents], seqs[n_agents * max_per_agent]) — one
@@ -1606,13 +1640,46 @@ public:
       HIP_CHECK(hipMemcpyAsync(d_needle_, nd.data(), nd.size(),
                                hipMemcpyHostToDevice, stream_));
       HIP_CHECK(hipMemsetAsync(d_match_count_, 0, sizeof(u32), stream_));
+      // the slot region of [lo, hi) is at most two contiguous byte
+      // ranges (ring wraps once): launch a pure linear-scan kernel per
+      // segment. Pipeline depth / grid size A/B'd on hardware (see
+      // profiles/ and csrc/bench_membw.hip).
+      static const int sP = [] {
+        const char *e = getenv("SWARMQ_SEARCH_P");
+        return e ? atoi(e) : 8;
+      }();
+      static const int sblocks = [] {
+        const char *e = getenv("SWARMQ_SEARCH_BLOCKS");
+        return e ? atoi(e) : 16384;
+      }();
+      const u32 cps = g_.slot_bytes >> 4;
       const u64 span = hi - lo;
-      const u64 chunks = span * (g_.slot_bytes >> 4);
-      const int blocks = (int)std::min<u64>((chunks + 255) / 256, 4096);
-      hipLaunchKernelGGL(k_search, dim3(blocks), dim3(256), 0, stream_, lo, hi,
-                         d_needle_, (int)nd.size(), fold ? 1 : 0, d_hdr_,
-                         d_status_, d_payload_, d_match_, d_match_count_, cap,
-                         g_);
+      const u32 slot_lo = (u32)(lo % g_.num_slots);
+      const u64 first = std::min<u64>(span, g_.num_slots - slot_lo);
+      struct Seg {
+        u64 off_slots, nslots, seq0;
+        u32 slot0;
+      } segs[2] = {{slot_lo, first, lo, slot_lo},
+                   {0, span - first, lo + first, 0}};
+      for (const Seg &s : segs) {
+        if (s.nslots == 0)
+          continue;
+        const u64 chunks = s.nslots * cps;
+        const int blocks =
+            (int)std::min<u64>((chunks + 255) / 256, sblocks);
+        auto kfn = k_search<8>;
+        if (sP == 4)
+          kfn = k_search<4>;
+        else if (sP == 2)
+          kfn = k_search<2>;
+        else if (sP == 16)
+          kfn = k_search<16>;
+        hipLaunchKernelGGL(kfn, dim3(blocks), dim3(256), 0, stream_,
+                           d_payload_ + s.off_slots * g_.slot_bytes, chunks,
+                           s.seq0, s.slot0, d_needle_, (int)nd.size(),
+                           fold ? 1 : 0, d_hdr_, d_status_, d_payload_,
+                           d_match_, d_match_count_, cap, g_);
+      }
       HIP_CHECK(hipMemcpyAsync(&nmatch, d_match_count_, sizeof(u32),
                                hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipStreamSynchronize(stream_));
@@ -2193,6 +2260,7 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("run_tick", &DeviceQueue::run_tick)
       .def_static("alloc_pinned", &DeviceQueue::alloc_pinned)
       .def("alloc_bitmap", &DeviceQueue::alloc_bitmap)
+      .def("get_bitmap", &DeviceQueue::get_bitmap)
       .def("pack_exchange", &DeviceQueue::pack_exchange)
       .def("enqueue_from_ptrs", &DeviceQueue::enqueue_from_ptrs)
       .def("receive_many", &DeviceQueue::receive_many, py::arg("agents"),

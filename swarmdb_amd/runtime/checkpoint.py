@@ -1,0 +1,307 @@
+"""Binary incremental checkpoints — the MI355X-native persistence plane.
+
+The reference's only persistence is a full-state JSON snapshot built one
+Python dict per message (reference swarmdb/ main.py:852-892) — fine at
+its scale, ~74 k msg/s against this engine's multi-million-per-second
+log. This module adds a BINARY checkpoint format that moves the log at
+device-gather speed (batched pinned D2H + vectorized numpy compaction,
+zero per-message Python), plus append-only DELTA segments between full
+snapshots (VERDICT round-1 item 8). The reference-schema JSON snapshot
+(`save_message_history`) remains the interchange format; this is the
+operational one.
+
+File layout — one or more self-contained segments, each:
+
+    magic   8 B   b"SWQCKPT1"
+    header  24 B  u64 meta_len | u64 n_records | u64 blob_len
+    meta    JSON  agents (idx order), registered, groups, llm
+                  assignments, agent_metadata, seq range, bitmap table,
+                  overflow directory
+    recs    n x 48 B REC_DTYPE (payload_off rebased to 16-B-aligned
+                  offsets into this segment's blob)
+    status  n x 1 B
+    blob    blob_len B (16-B-padded payloads, concatenated)
+
+``save_checkpoint`` writes one full segment (all retained seqs);
+``save_checkpoint_delta`` appends a segment with only the seqs created
+since the previous save to ``<base>.delta``; ``load_checkpoint`` replays
+base + delta segments with ONE engine enqueue + ONE status restore per
+segment. Visibility bitmaps referenced by saved records are persisted
+from the engine pool and re-allocated on load (records whose bitmap was
+already recycled at save time — i.e. already hidden — are pointed at an
+empty bitmap so they stay hidden).
+"""
+
+from __future__ import annotations
+
+import json
+import struct
+from pathlib import Path
+from typing import Dict, List, Optional, Tuple, Union
+
+import numpy as np
+
+from .engine import (
+    NO_BITMAP,
+    REC_DTYPE,
+    ST_DELETED,
+    VIS_ALL,
+)
+
+MAGIC = b"SWQCKPT1"
+_HDR = struct.Struct("<QQQ")
+CHUNK = 1 << 16
+
+
+def _segment_for(db, seq_lo: int, seq_hi: int) -> Tuple[bytes, int]:
+    """Serialize seqs [seq_lo, seq_hi) into one segment. Returns
+    (segment bytes, records written)."""
+    engine = db.engine
+    seq_lo = max(seq_lo, engine.evict_base())
+    seqs = np.arange(seq_lo, seq_hi, dtype=np.uint64)
+
+    rec_parts: List[np.ndarray] = []
+    status_parts: List[np.ndarray] = []
+    blob_parts: List[np.ndarray] = []
+    blob_off = 0
+    bitmap_dir: Dict[Tuple[int, int], Optional[list]] = {}
+    overflow_dir: List[list] = []
+
+    done = 0
+    written = 0
+    while done < len(seqs):
+        chunk = seqs[done : done + CHUNK]
+        hdrs, flat, stride = engine.fetch_raw_chunks(chunk)
+        alive = hdrs["status"] != ST_DELETED
+        hdrs = hdrs[alive]
+        rows = np.flatnonzero(alive)
+        if len(hdrs):
+            recs = np.zeros(len(hdrs), dtype=REC_DTYPE)
+            for name in REC_DTYPE.names:
+                recs[name] = hdrs[name]
+            lens = recs["payload_len"].astype(np.int64)
+            plen16 = (lens + 15) // 16 * 16
+            offs = np.zeros(len(recs), dtype=np.int64)
+            np.cumsum(plen16[:-1], out=offs[1:])
+            # ragged compaction: rows -> dense blob. Uniform lengths
+            # (the common case) are a plain 2-D slice copy; the mask
+            # path handles mixed sizes.
+            mat = flat.reshape(-1, stride)[rows]
+            L16 = int(plen16[0]) if len(plen16) else 0
+            if len(plen16) and (plen16 == L16).all():
+                dense = np.ascontiguousarray(mat[:, :L16]).reshape(-1)
+            else:
+                ar = np.arange(stride)
+                mask = ar[None, :] < plen16[:, None]
+                dense = mat[mask]  # concatenated 16-B-padded payloads
+            recs["payload_off"] = (offs + blob_off).astype(np.uint64)
+            blob_off += int(plen16.sum())
+            blob_parts.append(dense)
+            rec_parts.append(recs)
+            status_parts.append(hdrs["status"].astype(np.uint8))
+            # bitmap directory: persist each referenced (slot, epoch)
+            vis_rows = np.flatnonzero(
+                (recs["vis_mode"] != VIS_ALL) & (recs["bitmap"] != NO_BITMAP)
+            )
+            for i in vis_rows:
+                key = (int(recs["bitmap"][i]), int(recs["bitmap_epoch"][i]))
+                if key not in bitmap_dir:
+                    bits = engine.read_bitmap(*key)
+                    bitmap_dir[key] = (
+                        np.packbits(bits, bitorder="little")
+                        .tobytes()
+                        .hex()
+                        if bits is not None
+                        else None
+                    )
+            # host-side overflow payloads ride in the meta directory
+            for i, s in enumerate(hdrs["seq"]):
+                ov = db._overflow.get(int(s))
+                if ov is not None:
+                    clen, payload = ov
+                    overflow_dir.append(
+                        [written + i, int(clen), payload.hex()]
+                    )
+            written += len(hdrs)
+        done += CHUNK
+
+    recs = (
+        np.concatenate(rec_parts)
+        if rec_parts
+        else np.empty(0, dtype=REC_DTYPE)
+    )
+    statuses = (
+        np.concatenate(status_parts)
+        if status_parts
+        else np.empty(0, dtype=np.uint8)
+    )
+    blob = (
+        np.concatenate(blob_parts)
+        if blob_parts
+        else np.empty(0, dtype=np.uint8)
+    )
+
+    with db._lock:
+        meta = {
+            "version": 1,
+            "agents": list(db._agent_ids),
+            "registered": sorted(db.registered_agents),
+            "groups": db.metadata.get("agent_groups", {}),
+            "llm_backends": db.metadata.get("llm_backends", {}),
+            "agent_metadata": db.agent_metadata,
+            "seq_lo": int(seq_lo),
+            "seq_hi": int(seq_hi),
+            "slot_bytes": int(db.config.slot_bytes),
+            "bitmaps": {f"{k[0]}:{k[1]}": v for k, v in bitmap_dir.items()},
+            "overflow": overflow_dir,
+        }
+    meta_b = json.dumps(meta, separators=(",", ":")).encode()
+    seg = b"".join(
+        [
+            MAGIC,
+            _HDR.pack(len(meta_b), len(recs), len(blob)),
+            meta_b,
+            recs.tobytes(),
+            statuses.tobytes(),
+            blob.tobytes(),
+        ]
+    )
+    return seg, len(recs)
+
+
+def save_checkpoint(db, path: Union[str, Path, None] = None) -> str:
+    """Full binary snapshot of every retained message. Resets the delta
+    tracking point."""
+    from datetime import datetime
+
+    engine = db.engine
+    hi = engine.total_messages()
+    if path is None:
+        db.save_dir.mkdir(parents=True, exist_ok=True)
+        ts = datetime.now().strftime("%Y%m%d_%H%M%S")
+        path = db.save_dir / f"checkpoint_{ts}_{hi}.swq"
+    path = Path(path)
+    seg, n = _segment_for(db, engine.evict_base(), hi)
+    with open(path, "wb") as f:
+        f.write(seg)
+    # a fresh base invalidates any previous delta chain
+    delta = path.with_suffix(path.suffix + ".delta")
+    if delta.exists():
+        delta.unlink()
+    db._ckpt_base = str(path)
+    db._ckpt_high = hi
+    return str(path)
+
+
+def save_checkpoint_delta(db) -> Tuple[str, int]:
+    """Append messages created since the last save to the base's
+    ``.delta`` file (one self-contained segment per call)."""
+    base = getattr(db, "_ckpt_base", None)
+    if base is None:
+        raise RuntimeError("no base checkpoint: call save_checkpoint first")
+    engine = db.engine
+    hi = engine.total_messages()
+    lo = getattr(db, "_ckpt_high", 0)
+    delta_path = Path(base).with_suffix(Path(base).suffix + ".delta")
+    if hi <= lo:
+        return str(delta_path), 0
+    seg, n = _segment_for(db, lo, hi)
+    with open(delta_path, "ab") as f:
+        f.write(seg)
+    db._ckpt_high = hi
+    return str(delta_path), n
+
+
+def _load_segment(db, buf: memoryview, off: int) -> Tuple[int, int]:
+    """Replay one segment from buf[off:]. Returns (new offset, records
+    loaded)."""
+    if bytes(buf[off : off + 8]) != MAGIC:
+        raise ValueError("bad checkpoint magic")
+    off += 8
+    meta_len, n, blob_len = _HDR.unpack(buf[off : off + 24])
+    off += 24
+    meta = json.loads(bytes(buf[off : off + meta_len]))
+    off += meta_len
+    recs = np.frombuffer(buf, dtype=REC_DTYPE, count=n, offset=off).copy()
+    off += n * REC_DTYPE.itemsize
+    statuses = np.frombuffer(buf, dtype=np.uint8, count=n, offset=off).copy()
+    off += n
+    blob = np.frombuffer(buf, dtype=np.uint8, count=blob_len, offset=off)
+    off += blob_len
+
+    engine = db.engine
+    with db._lock:
+        # index remap: register the segment's agents (in ITS idx order
+        # first, so a fresh facade reproduces the dense table exactly)
+        remap = np.empty(max(len(meta["agents"]), 1), dtype=np.uint32)
+        for old_idx, agent in enumerate(meta["agents"]):
+            db.register_agent(agent)
+            remap[old_idx] = db._agent_idx[agent]
+        for agent in meta["registered"]:
+            db.register_agent(agent)
+        db.metadata.setdefault("agent_groups", {}).update(meta["groups"])
+        db.metadata.setdefault("llm_backends", {}).update(
+            meta["llm_backends"]
+        )
+        db.agent_metadata.update(meta["agent_metadata"])
+
+        if n == 0:
+            return off, 0
+        bcast = recs["receiver"] == np.uint32(0xFFFFFFFF)
+        recs["sender"] = remap[recs["sender"]]
+        recs["receiver"][~bcast] = remap[recs["receiver"][~bcast]]
+
+        # re-allocate the persisted visibility bitmaps in THIS engine
+        handle_map: Dict[str, int] = {}
+        empty_handle: Optional[int] = None
+        for key, words_hex in meta["bitmaps"].items():
+            if words_hex is None:
+                if empty_handle is None:
+                    empty_handle = engine.alloc_bitmap(
+                        np.zeros(db.config.max_agents, dtype=bool)
+                    )
+                handle_map[key] = empty_handle
+            else:
+                bits = np.unpackbits(
+                    np.frombuffer(bytes.fromhex(words_hex), dtype=np.uint8),
+                    bitorder="little",
+                ).astype(bool)
+                old_agents = meta["agents"]
+                nb = np.zeros(db.config.max_agents, dtype=bool)
+                set_old = np.flatnonzero(bits[: len(old_agents)])
+                nb[remap[set_old]] = True
+                handle_map[key] = engine.alloc_bitmap(nb)
+        vis_rows = np.flatnonzero(
+            (recs["vis_mode"] != VIS_ALL) & (recs["bitmap"] != NO_BITMAP)
+        )
+        for i in vis_rows:
+            key = f"{int(recs['bitmap'][i])}:{int(recs['bitmap_epoch'][i])}"
+            recs["bitmap"][i] = handle_map[key]
+            recs["bitmap_epoch"][i] = 0  # engine re-splits the handle
+
+        seqs = engine.enqueue_batch(recs, blob.tobytes())
+        engine.set_statuses(seqs, statuses.astype(np.uint32))
+        for idx, clen, payload_hex in meta["overflow"]:
+            db._overflow[int(seqs[idx])] = (clen, bytes.fromhex(payload_hex))
+    return off, n
+
+
+def load_checkpoint(db, path: Union[str, Path],
+                    with_deltas: bool = True) -> int:
+    """Replay a base checkpoint (and its delta chain) into ``db``.
+    Returns total records loaded."""
+    path = Path(path)
+    total = 0
+    for p in [path] + (
+        [path.with_suffix(path.suffix + ".delta")] if with_deltas else []
+    ):
+        if not p.exists():
+            continue
+        data = memoryview(p.read_bytes())
+        off = 0
+        while off < len(data):
+            off, n = _load_segment(db, data, off)
+            total += n
+    db._ckpt_base = str(path)
+    db._ckpt_high = db.engine.total_messages()
+    return total

@@ -186,6 +186,26 @@ class Engine(abc.ABC):
         """Return (headers REC-like structured array incl. status + seq,
         payload bytes list) for the given seqs."""
 
+    def fetch_raw_chunks(
+        self, seqs: np.ndarray
+    ) -> Tuple[np.ndarray, np.ndarray, int]:
+        """Bulk fetch WITHOUT per-message Python objects: returns
+        (headers structured array incl. status+seq, flat uint8 payload
+        array laid out [n, stride], stride). The binary checkpoint path
+        uses this to move the log at device-gather speed."""
+        hdrs, pays = self.fetch(seqs)
+        stride = int(self.cfg.slot_bytes)  # type: ignore[attr-defined]
+        flat = np.zeros((len(seqs), stride), dtype=np.uint8)
+        for i, p in enumerate(pays):
+            flat[i, : len(p)] = np.frombuffer(p, dtype=np.uint8)
+        return hdrs, flat.reshape(-1), stride
+
+    def read_bitmap(self, slot: int, epoch: int) -> Optional[np.ndarray]:
+        """Bit array of a visibility-bitmap pool slot if the epoch still
+        matches (i.e. the referencing record's visibility set is still
+        live); None when the slot was recycled."""
+        return None
+
     @abc.abstractmethod
     def set_status(self, seq: int, status: int) -> None: ...
 

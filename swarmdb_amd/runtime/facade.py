@@ -1127,6 +1127,32 @@ class SwarmsDB:
                 )
             return n
 
+    # ---- binary incremental checkpoints (operational persistence;
+    # the JSON history above stays the reference-compatible format) ----
+
+    def save_checkpoint(self, path: Optional[str] = None) -> str:
+        """Full binary snapshot at device-gather speed (batched pinned
+        D2H, zero per-message Python — see runtime/checkpoint.py).
+        Resets the delta chain."""
+        from .checkpoint import save_checkpoint
+
+        return save_checkpoint(self, path)
+
+    def save_checkpoint_delta(self):
+        """Append messages since the last checkpoint/delta to the
+        base's .delta file. Returns (path, records appended)."""
+        from .checkpoint import save_checkpoint_delta
+
+        return save_checkpoint_delta(self)
+
+    def load_checkpoint(self, path: Union[str, Path],
+                        with_deltas: bool = True) -> int:
+        """Replay a binary checkpoint (+ delta chain). Returns records
+        loaded."""
+        from .checkpoint import load_checkpoint
+
+        return load_checkpoint(self, path, with_deltas)
+
     def export_as_yaml(self) -> str:
         """History object as YAML (reference swarmdb/ main.py:936-971)."""
         history = self._history_object()

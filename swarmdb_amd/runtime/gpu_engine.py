@@ -254,6 +254,39 @@ class GpuEngine(Engine):
             done += chunk
         return out, pays
 
+    def fetch_raw_chunks(self, seqs: np.ndarray):
+        """Device gather + one pinned D2H per chunk; no per-message
+        Python objects (the binary-checkpoint hot path)."""
+        seqs = np.ascontiguousarray(seqs, dtype=np.uint64)
+        n = len(seqs)
+        stride = self._slot_bytes
+        out = np.zeros(n, dtype=_FETCH_DTYPE)
+        flat = np.empty((n, stride), dtype=np.uint8)
+        done = 0
+        while done < n:
+            chunk = min(self._staging, n - done)
+            sub = seqs[done : done + chunk]
+            hdr_b, status, pay_b = self.q.fetch(sub)
+            hdr = np.frombuffer(hdr_b, dtype=REC_DTYPE, count=chunk)
+            for name in REC_DTYPE.names:
+                out[name][done : done + chunk] = hdr[name]
+            out["status"][done : done + chunk] = status.astype(np.uint8)
+            out["seq"][done : done + chunk] = sub
+            flat[done : done + chunk] = np.frombuffer(
+                pay_b, dtype=np.uint8, count=chunk * stride
+            ).reshape(chunk, stride)
+            done += chunk
+        return out, flat.reshape(-1), stride
+
+    def read_bitmap(self, slot: int, epoch: int):
+        ep, words = self.q.get_bitmap(int(slot))
+        if int(ep) != int(epoch):
+            return None
+        bits = np.unpackbits(
+            np.frombuffer(words, dtype=np.uint8), bitorder="little"
+        )
+        return bits.astype(bool)[: self.cfg.max_agents]
+
     def deliver_payloads(
         self, seqs: np.ndarray, max_payload: int = 0, synchronize: bool = True
     ) -> int:
