@@ -407,8 +407,11 @@ def config7_checkpoint(seconds: float) -> None:
 
     from swarmdb_amd.runtime.gpu_engine import GpuEngine
 
+    import os
+
     tmp = tempfile.mkdtemp(prefix="swarmdb_ckpt_")
-    n_msgs = 1 << 17  # 128k x 256B messages
+    # 128k default; SWARMQ_CKPT_MSGS=1048576 for the 1M-message run
+    n_msgs = int(os.environ.get("SWARMQ_CKPT_MSGS", str(1 << 17)))
     cfg = QueueConfig(use_gpu=_gpu_available(), auto_save=False,
                       max_agents=256, num_slots=n_msgs,
                       slot_bytes=512, staging_batch=1 << 14,

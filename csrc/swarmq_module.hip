@@ -573,13 +573,17 @@ __device__ __forceinline__ void verify_chunk(
 // Linear-streaming scan over ONE contiguous slot segment. The slot
 // region of a seq range is always at most TWO contiguous byte ranges
 // (the ring wraps at most once), so the host splits the scan and each
-// launch is a PURE linear walk: addr = seg + 16*ci — the exact access
-// shape that measures 6.2 TB/s on this chip (csrc/bench_membw.hip),
-// where a per-chunk (slot, sub) walker with 64-bit address math topped
-// out at 2.9 TB/s. Headers/status load only for chunks that pass the
-// packed first-byte + second-byte screen (~P(char)^2 of chunks). Slot
-// padding bytes are scanned too and rejected by the content_len bound
-// at verify time.
+// launch is a PURE linear walk: addr = seg + 16*ci — the access shape
+// that measures 6.2 TB/s on this chip (csrc/bench_membw.hip).
+//
+// Cross-group software pipeline: group k+1's P loads are ISSUED before
+// group k's screens/verifies run, so the screen work (a tiered
+// first-char / adjacent-pair filter whose cost the wave pays whenever
+// any lane has a candidate, ~16% per lane on random text) overlaps the
+// loads' HBM round-trip instead of serializing against it. A
+// wave-aggregated two-phase variant (ballot-recorded candidates,
+// lane-parallel verify) measured WORSE (2.0 TB/s) — its per-ballot
+// phase-B overhead exceeded the screen it saved; see profiles/.
 template <int P>
 __global__ void k_search(const u8 *__restrict__ seg, u64 nchunks, u64 seq0,
                          u32 slot0, const u8 *__restrict__ needle, int nlen,
@@ -627,6 +631,9 @@ __global__ void k_search(const u8 *__restrict__ seg, u64 nchunks, u64 seq0,
       return false;
     if (nlen < 2)
       return true;
+    // tier 1.5 (issued at the first-char rate, ~16%): does the SECOND
+    // char appear at all — 4 more haszero tests gate the pricier
+    // positional tier down to ~P(char)^2 issue rate
     u32 f0 = byte_eq_mask(v.x, t1), f1 = byte_eq_mask(v.y, t1),
         f2 = byte_eq_mask(v.z, t1), f3 = byte_eq_mask(v.w, t1);
     if (two2) {
@@ -635,6 +642,8 @@ __global__ void k_search(const u8 *__restrict__ seg, u64 nchunks, u64 seq0,
       f2 |= byte_eq_mask(v.z, t2);
       f3 |= byte_eq_mask(v.w, t2);
     }
+    if ((f0 | f1 | f2 | f3) == 0)
+      return (e3 & 0x80000000u) != 0; // last-byte hit: pair spans chunks
     const u32 m0 = movemask4(e0) | (movemask4(e1) << 4) |
                    (movemask4(e2) << 8) | (movemask4(e3) << 12);
     const u32 m1 = movemask4(f0) | (movemask4(f1) << 4) |
@@ -653,21 +662,39 @@ __global__ void k_search(const u8 *__restrict__ seg, u64 nchunks, u64 seq0,
 
   const u64 stride = (u64)gridDim.x * blockDim.x;
   const uint4 *src = reinterpret_cast<const uint4 *>(seg);
+  const u64 group = (u64)P * stride;
   u64 ci = (u64)blockIdx.x * blockDim.x + threadIdx.x;
-  for (; ci + (u64)(P - 1) * stride < nchunks; ci += (u64)P * stride) {
-    uint4 v[P];
+
+  uint4 v[P];
+  bool have = ci + (u64)(P - 1) * stride < nchunks;
+  if (have) {
 #pragma unroll
-    for (int p = 0; p < P; ++p) // independent loads, all in flight
+    for (int p = 0; p < P; ++p)
       v[p] = src[ci + (u64)p * stride];
+  }
+  while (have) {
+    const u64 nci = ci + group;
+    const bool next = nci + (u64)(P - 1) * stride < nchunks;
+    uint4 vn[P];
+    if (next) {
+#pragma unroll
+      for (int p = 0; p < P; ++p) // next group's loads fly NOW,
+        vn[p] = src[nci + (u64)p * stride]; // over this group's screens
+    }
 #pragma unroll
     for (int p = 0; p < P; ++p)
       if (test16(v[p]))
         verify(ci + (u64)p * stride, v[p]);
+#pragma unroll
+    for (int p = 0; p < P; ++p)
+      v[p] = vn[p];
+    ci = nci;
+    have = next;
   }
   for (; ci < nchunks; ci += stride) {
-    const uint4 v = src[ci];
-    if (test16(v))
-      verify(ci, v);
+    const uint4 w = src[ci];
+    if (test16(w))
+      verify(ci, w);
   }
 }
 
@@ -1642,12 +1669,8 @@ public:
       HIP_CHECK(hipMemsetAsync(d_match_count_, 0, sizeof(u32), stream_));
       // the slot region of [lo, hi) is at most two contiguous byte
       // ranges (ring wraps once): launch a pure linear-scan kernel per
-      // segment. Pipeline depth / grid size A/B'd on hardware (see
-      // profiles/ and csrc/bench_membw.hip).
-      static const int sP = [] {
-        const char *e = getenv("SWARMQ_SEARCH_P");
-        return e ? atoi(e) : 8;
-      }();
+      // segment. Grid size A/B'd on hardware (see profiles/ and
+      // csrc/bench_membw.hip).
       static const int sblocks = [] {
         const char *e = getenv("SWARMQ_SEARCH_BLOCKS");
         return e ? atoi(e) : 16384;
@@ -1667,13 +1690,15 @@ public:
         const u64 chunks = s.nslots * cps;
         const int blocks =
             (int)std::min<u64>((chunks + 255) / 256, sblocks);
-        auto kfn = k_search<8>;
+        static const int sP = [] {
+          const char *e = getenv("SWARMQ_SEARCH_P");
+          return e ? atoi(e) : 2; // hardware A/B winner (profiles/)
+        }();
+        auto kfn = k_search<2>;
         if (sP == 4)
           kfn = k_search<4>;
-        else if (sP == 2)
-          kfn = k_search<2>;
-        else if (sP == 16)
-          kfn = k_search<16>;
+        else if (sP == 8)
+          kfn = k_search<8>;
         hipLaunchKernelGGL(kfn, dim3(blocks), dim3(256), 0, stream_,
                            d_payload_ + s.off_slots * g_.slot_bytes, chunks,
                            s.seq0, s.slot0, d_needle_, (int)nd.size(),
