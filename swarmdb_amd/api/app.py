@@ -102,6 +102,7 @@ def create_app(
     settings: Optional[ApiSettings] = None,
     credential_validator: Optional[Callable[[str, str], bool]] = None,
     batch_window: Optional[float] = None,
+    peer_urls: Optional[List[str]] = None,
 ) -> FastAPI:
     """App factory. ``db`` defaults to a SwarmsDB built from env config
     (GPU engine if a device is visible, CPU otherwise). One shared
@@ -206,6 +207,33 @@ def create_app(
     def is_admin(agent: str) -> bool:
         # authorization model: sub == "admin" (reference api.py:398 etc.)
         return agent == "admin"
+
+    def owner_redirect(agent_id: str, request: Request):
+        """Distributed REST gateway (round-1 VERDICT item 10): when this
+        process serves one rank of a DistributedSwarmsDB and the agent
+        lives on another rank, 307-redirect the request to the owner's
+        server (method + body preserved) instead of raising — in the
+        reference any worker serves any agent because Kafka is shared
+        (gunicorn_config.py:25-34); here the inbox is owner-local, so
+        the gateway routes the READ to the owner. Returns None when the
+        agent is local/unknown or no peer table is configured."""
+        if peer_urls is None:
+            return None
+        is_local = getattr(db, "is_local", None)
+        if is_local is None:
+            return None
+        try:
+            if is_local(agent_id):
+                return None
+            rank = db.owner_rank(agent_id)  # type: ignore[attr-defined]
+        except KeyError:
+            return None  # not registered yet: handle locally
+        from fastapi.responses import RedirectResponse
+
+        url = peer_urls[rank].rstrip("/") + request.url.path
+        if request.url.query:
+            url += "?" + request.url.query
+        return RedirectResponse(url, status_code=http.HTTP_307_TEMPORARY_REDIRECT)
 
     # ---------------- auth ----------------
 
@@ -499,31 +527,41 @@ def create_app(
             ]
         return [MessageResponse.from_message(m) for m in msgs]
 
-    @app.get("/agents/{agent_id}/messages", response_model=List[MessageResponse])
+    @app.get("/agents/{agent_id}/messages", response_model=None)
     async def get_agent_messages(
         agent_id: str,
+        request: Request,
         status: Optional[MessageStatus] = Query(None),
         limit: int = Query(100, ge=1, le=1000),
         skip: int = Query(0, ge=0),
         current: str = Depends(get_current_agent),
     ):
-        """reference api.py:624-664 (self-or-admin)."""
+        """reference api.py:624-664 (self-or-admin); cross-rank reads
+        are 307-redirected to the owner rank."""
         if agent_id != current and not is_admin(current):
             raise HTTPException(
                 status_code=http.HTTP_403_FORBIDDEN,
                 detail="Can only view your own messages",
             )
+        redir = owner_redirect(agent_id, request)
+        if redir is not None:
+            return redir
         msgs = db.get_agent_messages(agent_id, status=status, limit=limit, skip=skip)
         return [MessageResponse.from_message(m) for m in msgs]
 
-    @app.post("/agents/receive", response_model=List[MessageResponse])
+    @app.post("/agents/receive", response_model=None)
     async def receive_messages(
+        request: Request,
         max_messages: int = Query(100, ge=1, le=10000),
         timeout: float = Query(1.0, ge=0.0, le=30.0),
         priority_order: bool = Query(False),
         current: str = Depends(get_current_agent),
     ):
-        """The consumer-poll endpoint (reference api.py:667-688)."""
+        """The consumer-poll endpoint (reference api.py:667-688).
+        Cross-rank consumers are 307-redirected to their owner rank."""
+        redir = owner_redirect(current, request)
+        if redir is not None:
+            return redir
         msgs = db.receive_messages(
             current, max_messages=max_messages, timeout=timeout,
             priority_order=priority_order,
@@ -670,9 +708,10 @@ def create_app(
             )
         return SystemStats(**db.get_stats())
 
-    @app.get("/agents/{agent_id}/load", response_model=AgentLoadResponse)
+    @app.get("/agents/{agent_id}/load", response_model=None)
     async def agent_load(
-        agent_id: str, current: str = Depends(get_current_agent)
+        agent_id: str, request: Request,
+        current: str = Depends(get_current_agent),
     ):
         """get_agent_load, unrouted in the reference (swarmdb/
         main.py:1049-1094) — self-or-admin."""
@@ -680,11 +719,15 @@ def create_app(
             raise HTTPException(
                 status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
             )
+        redir = owner_redirect(agent_id, request)
+        if redir is not None:
+            return redir
         return AgentLoadResponse(**db.get_agent_load(agent_id))
 
     @app.get("/agents/{agent_id}/unread_count")
     async def unread_count(
-        agent_id: str, current: str = Depends(get_current_agent)
+        agent_id: str, request: Request,
+        current: str = Depends(get_current_agent),
     ):
         """get_unread_message_count, unrouted in the reference
         (swarmdb/ main.py:1026-1047)."""
@@ -692,6 +735,9 @@ def create_app(
             raise HTTPException(
                 status_code=http.HTTP_403_FORBIDDEN, detail="Admin only"
             )
+        redir = owner_redirect(agent_id, request)
+        if redir is not None:
+            return redir
         return {"agent_id": agent_id,
                 "unread_count": db.get_unread_message_count(agent_id)}
 

@@ -424,10 +424,17 @@ class DistributedSwarmsDB(SwarmsDB):
         timeout: float = 1.0,
         priority_order: bool = False,
     ):
+        if agent_id not in self._agent_idx:
+            # unknown agent: queue the registration (applied at the next
+            # tick, reference auto-register semantics) — nothing to read
+            self.register_agent(agent_id)
+            return []
         if not self.is_local(agent_id):
             raise RuntimeError(
                 f"agent '{agent_id}' lives on rank "
-                f"{self.owner_rank(agent_id)}; poll there"
+                f"{self.owner_rank(agent_id)}; poll there — or serve "
+                "through the REST gateway (create_app(peer_urls=...)), "
+                "which 307-redirects to the owner"
             )
         with self._lock:
             idx = self._agent_idx.get(agent_id)
