@@ -66,7 +66,8 @@ class GpuDirectRouter:
         # read them (2 ticks deep)
         self._hold: list = []
 
-    def route_and_enqueue(self, recs: np.ndarray, payloads: bytes) -> int:
+    def route_and_enqueue(self, recs: np.ndarray, payloads: bytes,
+                          owner_of: Optional[np.ndarray] = None) -> int:
         W = self.world
         q = self.engine.q
         if W == 1 and not self.force_exchange:
@@ -75,7 +76,14 @@ class GpuDirectRouter:
 
         recv_field = recs["receiver"]
         bmask = recv_field == BROADCAST
-        dest = np.where(bmask, 0, recv_field % np.uint32(W)).astype(np.int64)
+        if owner_of is None:
+            dest = np.where(
+                bmask, 0, recv_field % np.uint32(W)
+            ).astype(np.int64)
+        else:
+            dest = np.where(
+                bmask, 0, owner_of[np.where(bmask, 0, recv_field)]
+            ).astype(np.int64)
 
         # instance table: one per p2p message, W per broadcast
         p2p_idx = np.flatnonzero(~bmask)
@@ -197,17 +205,25 @@ class CrossGpuRouter:
         self.rank = dist.get_rank(group)
 
     def route(
-        self, recs: np.ndarray, payloads: bytes
+        self, recs: np.ndarray, payloads: bytes,
+        owner_of: Optional[np.ndarray] = None,
     ) -> Tuple[np.ndarray, bytes]:
         """Partition by destination, exchange, return this rank's inbound
         (records, payloads) — remote plus own, payload offsets rebased to
-        the returned buffer."""
+        the returned buffer. ``owner_of`` (agent idx -> rank, replicated)
+        overrides the default modulo sharding — live re-sharding routes
+        through it."""
         W = self.world
         recv = recs["receiver"]
-        dest = np.where(
-            recv == BROADCAST, self.rank, recv % np.uint32(W)
-        ).astype(np.int64)
         bmask = recv == BROADCAST
+        if owner_of is None:
+            dest = np.where(
+                bmask, self.rank, recv % np.uint32(W)
+            ).astype(np.int64)
+        else:
+            dest = np.where(
+                bmask, self.rank, owner_of[np.where(bmask, 0, recv)]
+            ).astype(np.int64)
 
         # build per-destination chunks (broadcasts replicated to all)
         src = np.frombuffer(payloads, dtype=np.uint8)

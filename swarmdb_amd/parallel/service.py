@@ -84,6 +84,9 @@ class DistributedSwarmsDB(SwarmsDB):
             else torch.device("cpu")
         )
         self._router = CrossGpuRouter(device, group)
+        # replicated ownership table (agent idx -> rank); default is
+        # idx % world, rewritten by live migration ops
+        self._owner_vec = np.full(config.max_agents, -1, dtype=np.int64)
         # control-plane tensors live wherever the collective backend
         # wants them (cuda for nccl/RCCL, cpu for gloo)
         self._ctl_device = device
@@ -95,16 +98,16 @@ class DistributedSwarmsDB(SwarmsDB):
     # ------------------------------------------------------------------
 
     def owner_rank(self, agent_id: str) -> int:
-        """Owner = dense index mod world — the same mapping the
-        all-to-all router applies to records, and deterministic across
-        ranks because index assignment is tick-synchronized. Requires the
+        """Owner rank from the replicated ownership table (defaults to
+        dense index mod world; live migration rewrites entries). The
+        table is tick-synchronized, so every rank agrees. Requires the
         agent to be registered (and a tick to have run)."""
         idx = self._agent_idx.get(agent_id)
         if idx is None:
             raise KeyError(
                 f"agent '{agent_id}' not registered yet (register + tick)"
             )
-        return idx % self.world
+        return int(self._owner_vec[idx])
 
     def is_local(self, agent_id: str) -> bool:
         return self.owner_rank(agent_id) == self.rank
@@ -132,6 +135,8 @@ class DistributedSwarmsDB(SwarmsDB):
             if agent_id in self.registered_agents:
                 return
             idx = self._idx_of(agent_id, create=True)
+            if self._owner_vec[idx] < 0:
+                self._owner_vec[idx] = idx % self.world
             # active (inbox fan-out target) only on the owner rank; the
             # index table itself is replicated on every rank
             if self.is_local(agent_id):
@@ -150,6 +155,75 @@ class DistributedSwarmsDB(SwarmsDB):
             idx = self._agent_idx[agent_id]
             if self.is_local(agent_id):
                 self.engine.deregister_agent(idx)
+
+    def migrate_agent(self, agent_id: str, new_rank: int) -> None:
+        """Queue a live re-shard of one agent to ``new_rank`` (the
+        re-sharding the reference approximates with Kafka partition
+        growth, swarmdb/ main.py:1327-1365 — here ownership actually
+        moves). Applied at the next tick on every rank: the old owner
+        drains the agent's undelivered inbox and re-homes those
+        messages through the same exchange, the new owner activates the
+        inbox, and the replicated ownership table flips — in-flight and
+        later sends route to the new rank. The agent's already-read
+        HISTORY stays on the old shard (shard-local queries, as
+        documented)."""
+        if not (0 <= new_rank < self.world):
+            raise ValueError(f"rank {new_rank} outside world {self.world}")
+        self._queue_ctl(("migrate", agent_id, int(new_rank)))
+
+    def rebalance_plan(self) -> Dict[str, int]:
+        """Round-robin re-spread of all registered agents (a helper for
+        operators; apply with migrate_agent + tick)."""
+        with self._lock:
+            agents = sorted(self.registered_agents)
+        return {a: i % self.world for i, a in enumerate(agents)}
+
+    def _apply_migrate(self, agent_id: str, new_rank: int) -> None:
+        with self._lock:
+            idx = self._agent_idx.get(agent_id)
+            if idx is None:
+                return
+            old = int(self._owner_vec[idx])
+            if old == new_rank:
+                return
+            self._owner_vec[idx] = new_rank
+            if new_rank == self.rank:
+                self.engine.register_agent(idx)
+            if old != self.rank:
+                return
+            # old owner: drain undelivered entries and re-home them
+            # through this tick's exchange (they route by the updated
+            # ownership table)
+            while True:
+                seqs = self.engine.receive(idx, 4096)
+                if len(seqs) == 0:
+                    break
+                hdrs, flat, stride = self.engine.fetch_raw_chunks(seqs)
+                recs = np.zeros(len(hdrs), dtype=REC_DTYPE)
+                for name in REC_DTYPE.names:
+                    recs[name] = hdrs[name]
+                recs["receiver"] = idx  # re-home as direct deliveries
+                unsplit = getattr(self.engine, "unsplit_bitmap_handles",
+                                  None)
+                if unsplit is not None:
+                    recs = unsplit(recs)
+                lens = recs["payload_len"].astype(np.int64)
+                plen16 = (lens + 15) // 16 * 16
+                offs = np.zeros(len(recs), dtype=np.int64)
+                np.cumsum(plen16[:-1], out=offs[1:])
+                mat = flat.reshape(-1, stride)
+                ar = np.arange(stride)
+                mask = ar[None, :] < plen16[:, None]
+                payload = mat[mask].tobytes()
+                recs["payload_off"] = (
+                    offs + self._out_bytes
+                ).astype(np.uint64)
+                self._out_recs.append(recs)
+                self._out_pay.append(payload)
+                self._out_bytes += len(payload)
+            # deactivate locally (the agent stays registered
+            # service-wide; fan-out targets only the new owner's engine)
+            self.engine.deregister_agent(idx)
 
     def add_agent_group(self, group_name: str, agent_ids: List[str]) -> None:  # type: ignore[override]
         for a in agent_ids:
@@ -354,6 +428,8 @@ class DistributedSwarmsDB(SwarmsDB):
                 self._apply_deregister(args[0])
             elif kind == "group":
                 self._apply_group(args[0], args[1])
+            elif kind == "migrate":
+                self._apply_migrate(args[0], args[1])
         # replicated bitmap allocation (identical pool on every rank)
         my_bitmaps: List[int] = []
         for r in range(self.world):
@@ -369,8 +445,10 @@ class DistributedSwarmsDB(SwarmsDB):
                 msg, next(bit_iter) if msg.visible_to else None
             )
 
-        if not any_sends:
-            return 0  # quiet data plane: skip the router exchange
+        # control ops can generate handoff traffic (migration re-homing)
+        # even when no rank queued sends
+        if not any_sends and max(ctl_sizes) == 0:
+            return 0  # quiet tick: skip the router exchange
 
         if self._out_recs:
             recs = np.concatenate(self._out_recs)
@@ -380,7 +458,9 @@ class DistributedSwarmsDB(SwarmsDB):
             payload = b""
         self._out_recs, self._out_pay, self._out_bytes = [], [], 0
 
-        in_recs, in_pay = self._router.route(recs, payload)
+        in_recs, in_pay = self._router.route(
+            recs, payload, owner_of=self._owner_vec
+        )
         if len(in_recs):
             seqs = self.engine.enqueue_batch(in_recs, in_pay)
             # map ids for locally-ingested compat-path messages

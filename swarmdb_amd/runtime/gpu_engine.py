@@ -125,6 +125,20 @@ class GpuEngine(Engine):
         out["bitmap"][mask] = bm[mask] % np.uint32(self._num_bitmaps)
         return out
 
+    def unsplit_bitmap_handles(self, recs: np.ndarray) -> np.ndarray:
+        """Inverse of split_bitmap_handles for records FETCHED from the
+        device (e.g. migration handoff): restore the raw handle (the
+        epoch IS the allocation counter) so a later enqueue re-splits
+        it."""
+        bm = recs["bitmap"]
+        mask = (recs["vis_mode"] != 0) & (bm != 0xFFFFFFFF)
+        if not mask.any():
+            return recs
+        out = recs.copy()
+        out["bitmap"][mask] = out["bitmap_epoch"][mask]
+        out["bitmap_epoch"][mask] = 0
+        return out
+
     def enqueue_batch(self, recs: np.ndarray, payloads: bytes) -> np.ndarray:
         n = len(recs)
         if n == 0:
