@@ -496,3 +496,25 @@ def test_compat_host_maps_prune_with_retention(tmp_path):
     assert ids[0] not in db._id_to_seq
     assert db._id_to_seq.get(ids[-1]) is not None
     db.close()
+
+
+def test_express_lane_cpu_double(tmp_path):
+    """Express-lane API contract on the CPU double (the GPU path is the
+    doorbell persistent kernel, covered by the gpu-marked test)."""
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path), auto_save=False,
+                      max_agents=64)
+    db = SwarmsDB(config=cfg)
+    db.express_start(["a", "b", "c"])
+    assert db.express_recv("b", timeout_us=0) is None
+    db.express_send("a", "b", "fast one")
+    db.express_send("c", "b", b"\x00binary\xff")
+    s1 = db.express_recv("b")
+    s2 = db.express_recv("b")
+    assert s1 == ("a", b"fast one")
+    assert s2 == ("c", b"\x00binary\xff")
+    with pytest.raises(RuntimeError):
+        db.express_start(["x"])  # already running
+    db.express_stop()
+    with pytest.raises(RuntimeError):
+        db.express_send("a", "b", "nope")
+    db.close()

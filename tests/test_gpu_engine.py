@@ -858,3 +858,28 @@ def test_doorbell_express_latency():
         db.stop()
         assert db.exited() or not db.running()
         db.release()
+
+
+def test_express_lane_facade_gpu(tmp_path):
+    """Facade express lane over the real doorbell kernel."""
+    from swarmdb_amd import QueueConfig, SwarmsDB
+
+    cfg = QueueConfig(use_gpu=True, save_dir=str(tmp_path),
+                      auto_save=False, max_agents=256,
+                      num_slots=1 << 14, slot_bytes=512,
+                      inbox_capacity=1 << 12, staging_batch=4096)
+    db = SwarmsDB(config=cfg)
+    try:
+        db.express_start(["ping", "pong"], max_seconds=20.0)
+        lat = []
+        for i in range(100):
+            t0 = time.perf_counter()
+            db.express_send("ping", "pong", f"msg{i}")
+            got = db.express_recv("pong", timeout_us=2e6)
+            lat.append(time.perf_counter() - t0)
+            assert got == ("ping", f"msg{i}".encode()), got
+        p50 = float(np.median(lat) * 1e6)
+        print(f"\nfacade express p50={p50:.1f}us")
+        assert p50 < 1000.0
+    finally:
+        db.close()
