@@ -96,7 +96,7 @@ def main() -> int:
                          "(each fans out to every agent)")
     ap.add_argument("--no-gather", action="store_true",
                     help="skip payload D2H gather (delivery stays device-side)")
-    ap.add_argument("--sample-every", type=int, default=4,
+    ap.add_argument("--sample-every", type=int, default=8,
                     help="latency-sample every Nth step (drains the "
                          "delivery D2H on sampled steps)")
     ap.add_argument("--dump-steps", action="store_true",
