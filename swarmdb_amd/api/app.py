@@ -844,6 +844,52 @@ def create_app(
         _require_admin(current)
         return {"status": "exported", "path": db.export_as_yaml()}
 
+    @app.post("/admin/checkpoint")
+    async def admin_checkpoint(current: str = Depends(get_current_agent)):
+        """Binary full checkpoint (device-gather speed; the JSON
+        /admin/save stays the reference-compatible format)."""
+        _require_admin(current)
+        return {"status": "checkpointed", "path": db.save_checkpoint()}
+
+    @app.post("/admin/checkpoint/delta")
+    async def admin_checkpoint_delta(
+        current: str = Depends(get_current_agent),
+    ):
+        """Append-only delta since the last checkpoint/delta."""
+        _require_admin(current)
+        try:
+            path, n = db.save_checkpoint_delta()
+        except RuntimeError as e:
+            raise HTTPException(
+                status_code=http.HTTP_409_CONFLICT, detail=str(e)
+            ) from None
+        return {"status": "delta", "path": path, "messages": n}
+
+    @app.post("/admin/checkpoint/load")
+    async def admin_checkpoint_load(
+        path: str = Query(...), current: str = Depends(get_current_agent)
+    ):
+        """Replay a binary checkpoint (+ delta chain); path constrained
+        to the save directory like /admin/load."""
+        _require_admin(current)
+        from pathlib import Path as _Path
+
+        save_root = _Path(db.save_dir).resolve()
+        target = _Path(path).resolve()
+        if save_root not in target.parents and target != save_root:
+            raise HTTPException(
+                status_code=http.HTTP_403_FORBIDDEN,
+                detail="checkpoint path must live under the save directory",
+            )
+        try:
+            n = db.load_checkpoint(target)
+        except FileNotFoundError:
+            raise HTTPException(
+                status_code=http.HTTP_404_NOT_FOUND,
+                detail=f"Checkpoint not found: {path}",
+            ) from None
+        return {"status": "loaded", "messages": n}
+
     return app
 
 

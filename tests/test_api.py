@@ -532,3 +532,27 @@ def test_agent_messages_pagination_matches_reference_order(client):
     r = client.get("/agents/b/messages?status=processed&skip=1&limit=10",
                    headers=hb)
     assert r.json() == []
+
+
+def test_admin_checkpoint_routes(client):
+    ha = auth(client, "admin")
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "ckpt me"})
+    # no base yet -> delta conflicts
+    r = client.post("/admin/checkpoint/delta", headers=ha)
+    assert r.status_code == 409
+    r = client.post("/admin/checkpoint", headers=ha)
+    assert r.status_code == 200
+    path = r.json()["path"]
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "delta me"})
+    r = client.post("/admin/checkpoint/delta", headers=ha)
+    assert r.status_code == 200 and r.json()["messages"] == 1
+    r = client.post(f"/admin/checkpoint/load?path={path}", headers=ha)
+    assert r.status_code == 200 and r.json()["messages"] == 2
+    # path constraint mirrors /admin/load
+    r = client.post("/admin/checkpoint/load?path=/etc/passwd", headers=ha)
+    assert r.status_code == 403
+    # non-admin rejected
+    r = client.post("/admin/checkpoint", headers=auth(client, "a"))
+    assert r.status_code == 403
