@@ -12,11 +12,12 @@ Each timed step is one full delivery tick through the GPU-resident queue:
 Latency is measured PER MESSAGE, not per tick: every message header
 carries its submission timestamp (stamped when the batch is handed to
 the queue — for the pipelined path that is at pinned-buffer staging, so
-pipeline wait is included); on sampled steps the bench drains the
-delivery D2H, records the wall-clock receive time, and keeps a random
-sample of delivered seqs. After the timed loop it fetches those headers
-and reports p50/p99 of (receive time - submission timestamp). The median
-step duration is still reported as ms_per_step.
+pipeline wait is included). A dedicated steady-state latency phase runs
+immediately AFTER the timed loop (sampling drains the async delivery
+pipeline, ~6% throughput, which must not contaminate the timed region):
+each phase step drains the delivery D2H, records the wall-clock receive
+time and keeps a random sample of delivered seqs; p50/p99 of
+(receive time - submission timestamp) are computed from those headers.
 
 Usage (driver contract):
   python bench.py --gpus N --steps K --warmup W
@@ -96,9 +97,6 @@ def main() -> int:
                          "(each fans out to every agent)")
     ap.add_argument("--no-gather", action="store_true",
                     help="skip payload D2H gather (delivery stays device-side)")
-    ap.add_argument("--sample-every", type=int, default=8,
-                    help="latency-sample every Nth step (drains the "
-                         "delivery D2H on sampled steps)")
     ap.add_argument("--dump-steps", action="store_true",
                     help="print per-step wall times (variance diagnosis)")
     ap.add_argument("--graph", action="store_true",
@@ -201,11 +199,14 @@ def main() -> int:
     sent_total = 0
     recv_total = 0
 
-    # per-message latency sampling: on every `sample_every`-th step the
-    # delivery D2H is drained, the receive wall-time recorded, and a
-    # random subset of delivered seqs kept; deltas against the header
-    # timestamps are computed after the timed loop
-    sample_every = max(1, args.sample_every)
+    # per-message latency sampling: on sampled steps the delivery D2H
+    # is drained, the receive wall-time recorded, and a random subset
+    # of delivered seqs kept; deltas against the header timestamps are
+    # computed at the end. Sampling runs in a DEDICATED steady-state
+    # phase after the timed loop (draining the async delivery pipeline
+    # costs ~6% throughput, which must not contaminate the timed
+    # region); the latencies are still real per-message measurements.
+    sampling = {"on": False}
     lat_samples: list = []  # (t_recv, seqs ndarray)
 
     def keep_sample(t_recv: float, seqs, counts=None, K=None) -> None:
@@ -273,7 +274,7 @@ def main() -> int:
                 deliver(seqs, args.payload)
             else:
                 engine.fetch(seqs)
-        if i % sample_every == 0 and ndel:
+        if sampling["on"] and ndel:
             keep_sample(time.time(), seqs)
         sent_total += sent_local
         recv_total += ndel
@@ -330,7 +331,7 @@ def main() -> int:
             nonlocal sent_total, recv_total
             cur = _cur[0]
             n_staged = len(batches[i % len(batches)][0])
-            sampled = i % sample_every == 0
+            sampled = sampling["on"]
             seqs = None
             counts_dense = None
             if use_graph:
@@ -398,6 +399,20 @@ def main() -> int:
     barrier_sync()
     t1 = time.perf_counter()
 
+    # latency phase: same steady state, sampling on, UNTIMED (every
+    # rank runs the same count — the distributed step has collectives)
+    lat_steps = max(4, args.steps // 4)
+    sampling["on"] = True
+    lat_sent = lat_recv = 0
+    for i in range(lat_steps):
+        before_s, before_r = sent_total, recv_total
+        step(args.warmup + args.steps + i)
+        lat_sent += sent_total - before_s
+        lat_recv += recv_total - before_r
+    sampling["on"] = False
+    sent_total -= lat_sent  # latency phase is outside the timed totals
+    recv_total -= lat_recv
+
     elapsed = t1 - t0
     if dist_on:
         import torch.distributed as dist
@@ -450,7 +465,7 @@ def main() -> int:
             p50_ms = float(np.median(alld) * 1000.0)
             p99_ms = float(np.percentile(alld, 99) * 1000.0)
     if p50_ms is None:
-        # no samples (e.g. --sample-every > steps): fall back to the
+        # no samples (zero deliveries in the latency phase): fall back to
         # tick-duration bound and say so via n_lat_samples = 0
         p50_ms = float(np.median(step_times) * 1000.0)
         p99_ms = float(np.percentile(step_times, 99) * 1000.0)
