@@ -382,8 +382,32 @@ def main() -> int:
             _cur[0] = 1 - cur
             return ndel
 
-    for i in range(args.warmup):
-        step(i)
+    # Warmup: at least args.warmup steps AND at least 150 ms of wall
+    # time. The floor is clock/power stabilization: with short steps a
+    # one-time ~6.5 ms reclock event otherwise lands a few steps into
+    # the timed region and poisons the mean (measured; profiles/
+    # r02_results.md). Warmup work never counts toward the timed total.
+    it = 0
+    tw = time.perf_counter()
+    for _ in range(args.warmup):
+        step(it)
+        it += 1
+    elapsed_w = time.perf_counter() - tw
+    extra = 0
+    if elapsed_w < 0.15 and args.warmup > 0:
+        per = max(elapsed_w / args.warmup, 1e-5)
+        extra = int((0.15 - elapsed_w) / per) + 1
+    if dist_on:
+        # every rank must run the same number of steps (collectives):
+        # agree on the max
+        import torch.distributed as dist
+
+        t = torch.tensor([extra], dtype=torch.int64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        extra = int(t.item())
+    for _ in range(extra):
+        step(it)
+        it += 1
 
     # warmup deliveries must not count toward the timed total
     sent_total = 0
@@ -392,9 +416,10 @@ def main() -> int:
     barrier_sync()
     t0 = time.perf_counter()
     step_times = []
-    for i in range(args.steps):
+    for _ in range(args.steps):
         s = time.perf_counter()
-        step(args.warmup + i)
+        step(it)
+        it += 1
         step_times.append(time.perf_counter() - s)
     barrier_sync()
     t1 = time.perf_counter()
@@ -404,9 +429,10 @@ def main() -> int:
     lat_steps = max(4, args.steps // 4)
     sampling["on"] = True
     lat_sent = lat_recv = 0
-    for i in range(lat_steps):
+    for _ in range(lat_steps):
         before_s, before_r = sent_total, recv_total
-        step(args.warmup + args.steps + i)
+        step(it)
+        it += 1
         lat_sent += sent_total - before_s
         lat_recv += recv_total - before_r
     sampling["on"] = False
