@@ -883,3 +883,44 @@ def test_express_lane_facade_gpu(tmp_path):
         assert p50 < 1000.0
     finally:
         db.close()
+
+
+def test_binary_checkpoint_roundtrip_gpu(tmp_path):
+    """Binary base+delta checkpoint over the GPU engine: statuses,
+    restricted visibility (epoch-tagged bitmaps) and payloads replay
+    into a fresh GPU facade."""
+    from swarmdb_amd import QueueConfig, SwarmsDB
+
+    def cfg():
+        return QueueConfig(use_gpu=True, save_dir=str(tmp_path),
+                           auto_save=False, max_agents=256,
+                           num_slots=1 << 14, slot_bytes=512,
+                           inbox_capacity=1 << 12, staging_batch=4096)
+
+    db = SwarmsDB(config=cfg())
+    for a in ["alice", "bob", "carol"]:
+        db.register_agent(a)
+    m1 = db.send_message("alice", "plain", receiver_id="bob")
+    db.send_message("alice", "secret", receiver_id=None,
+                    visible_to=["carol"])
+    db.broadcast_message("bob", "to everyone")
+    db.mark_message_as_processed(m1)
+    base = db.save_checkpoint()
+    db.send_message("carol", "late", receiver_id="bob")
+    _, nd = db.save_checkpoint_delta()
+    assert nd == 1
+
+    db2 = SwarmsDB(config=cfg())
+    loaded = db2.load_checkpoint(base)
+    assert loaded == 4
+    got_bob = db2.receive_messages("bob", timeout=0)
+    assert sorted(m.content for m in got_bob) == ["late", "plain"]
+    got_carol = db2.receive_messages("carol", timeout=0)
+    assert sorted(m.content for m in got_carol) == ["secret", "to everyone"]
+    got_alice = db2.receive_messages("alice", timeout=0)
+    assert [m.content for m in got_alice] == ["to everyone"]
+    assert [m.content for m in db2.query_messages(status="processed")] == [
+        "plain"
+    ]
+    db.close()
+    db2.close()
