@@ -413,6 +413,30 @@ def main() -> int:
     sent_total = 0
     recv_total = 0
 
+    # best-effort GPU-busy sampler (the driver's own sampler can miss a
+    # short timed region entirely — round-1 weak item 9): read the
+    # sysfs gpu_busy_percent every 50 ms on a daemon thread
+    busy_samples: list = []
+    stop_busy = False
+    if have_gpu and rank == 0:
+        import glob as _glob
+        import threading as _threading
+
+        paths = _glob.glob(
+            "/sys/class/drm/card*/device/gpu_busy_percent"
+        )
+
+        def _sample_busy():
+            while not stop_busy and paths:
+                try:
+                    with open(paths[0]) as f:
+                        busy_samples.append(float(f.read().strip()))
+                except OSError:
+                    return
+                time.sleep(0.05)
+
+        _threading.Thread(target=_sample_busy, daemon=True).start()
+
     barrier_sync()
     t0 = time.perf_counter()
     step_times = []
@@ -423,6 +447,7 @@ def main() -> int:
         step_times.append(time.perf_counter() - s)
     barrier_sync()
     t1 = time.perf_counter()
+    stop_busy = True
 
     # latency phase: same steady state, sampling on, UNTIMED (every
     # rank runs the same count — the distributed step has collectives)
@@ -514,6 +539,11 @@ def main() -> int:
             "p50_latency_ms": round(p50_ms, 3),
             "p99_latency_ms": round(p99_ms, 3),
             "n_latency_samples": n_lat,
+            "gpu_busy_pct": (
+                round(float(np.mean(busy_samples)), 1)
+                if busy_samples
+                else None
+            ),
             "config": {
                 "model": "gpu-mpmc-ring-queue",
                 "global_batch": args.batch * world,
