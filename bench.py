@@ -539,9 +539,12 @@ def main() -> int:
             "p50_latency_ms": round(p50_ms, 3),
             "p99_latency_ms": round(p99_ms, 3),
             "n_latency_samples": n_lat,
+            # None when sysfs gpu_busy_percent is absent or pinned at 0
+            # (non-functional in this pool's containers — the rocprofv3
+            # kernel evidence in profiles/ covers utilization instead)
             "gpu_busy_pct": (
                 round(float(np.mean(busy_samples)), 1)
-                if busy_samples
+                if busy_samples and max(busy_samples) > 0
                 else None
             ),
             "config": {
