@@ -69,6 +69,14 @@ class SwarmDBClient:
         if r.status_code == 401:  # token expired: refresh once
             self.login()
             r = self._http.request(method, path, headers=self._headers, **kw)
+        if r.status_code == 307:
+            # distributed gateway: the agent lives on another rank —
+            # follow to its server, re-sending auth (httpx would strip
+            # Authorization on a cross-host redirect; the JWT is valid
+            # on every rank)
+            r = self._http.request(
+                method, r.headers["location"], headers=self._headers, **kw
+            )
         r.raise_for_status()
         return r.json()
 
@@ -186,6 +194,23 @@ class SwarmDBClient:
 
     def complete_llm(self, backend_id: str) -> dict:
         return self._req("POST", f"/llm/complete/{backend_id}")
+
+    # ---- admin persistence (admin token required) ----
+
+    def admin_save(self) -> dict:
+        """Reference-schema JSON snapshot."""
+        return self._req("POST", "/admin/save")
+
+    def admin_checkpoint(self) -> dict:
+        """Binary full checkpoint (device-gather speed)."""
+        return self._req("POST", "/admin/checkpoint")
+
+    def admin_checkpoint_delta(self) -> dict:
+        return self._req("POST", "/admin/checkpoint/delta")
+
+    def admin_checkpoint_load(self, path: str) -> dict:
+        return self._req("POST", "/admin/checkpoint/load",
+                         params={"path": path})
 
     # ---- probes ----
 

@@ -72,3 +72,16 @@ def test_client_end_to_end(served_client_factory):
     assert [m["id"] for m in mine] == [bid]
     q = alice.query(sender_id="alice", limit=3)
     assert len(q) == 3
+
+
+def test_client_admin_checkpoint(served_client_factory):
+    admin = served_client_factory("admin")
+    alice = served_client_factory("alice")
+    alice.register()
+    alice.send("admin", "persist me")
+    out = admin.admin_checkpoint()
+    assert out["status"] == "checkpointed"
+    alice.send("admin", "delta me")
+    assert admin.admin_checkpoint_delta()["messages"] == 1
+    loaded = admin.admin_checkpoint_load(out["path"])
+    assert loaded["messages"] == 2
