@@ -224,3 +224,98 @@ def test_rccl_gpu_direct_world2_single_device(tmp_path):
                     + out.split("RCCL_INIT_FAILED:")[1][:200])
     assert proc.returncode == 0, out
     assert '"ok": true' in out.lower()
+
+
+WORKER_SVC = textwrap.dedent(
+    """
+    import json, os, sys
+    sys.path.insert(0, %r)
+    import numpy as np
+    import torch
+    import torch.distributed as dist
+
+    from swarmdb_amd import QueueConfig
+    from swarmdb_amd.parallel.service import DistributedSwarmsDB
+
+    # gloo exchange (host buffers) + GPU-resident shards on ONE device:
+    # the service tier over real DeviceQueues without needing RCCL
+    # multi-rank-per-device support
+    dist.init_process_group(backend="gloo")
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    torch.cuda.set_device(0)
+
+    cfg = QueueConfig(use_gpu=True, auto_save=False, max_agents=64,
+                      num_slots=1 << 14, slot_bytes=512,
+                      inbox_capacity=1 << 12, staging_batch=4096)
+    svc = DistributedSwarmsDB(config=cfg)
+    agents = [f"agent{i}" for i in range(8)]
+    for a in agents:
+        svc.register_agent(a)
+    svc.tick(); svc.tick()
+
+    own = [a for a in agents if svc.is_local(a)]
+    ids = {}
+    for s in own:
+        for r_ in agents:
+            if r_ != s:
+                ids[(s, r_)] = svc.send_message(
+                    s, f"{s}->{r_}", receiver_id=r_)
+    svc.tick(); svc.tick()
+
+    got = 0
+    for a in own:
+        msgs = svc.receive_messages(a, max_messages=100, timeout=0)
+        for m in msgs:
+            assert m.content.endswith("->" + a), m.content
+        got += len(msgs)
+    t = torch.tensor([got], dtype=torch.int64)
+    dist.all_reduce(t)
+    expect = len(agents) * (len(agents) - 1)
+    assert int(t.item()) == expect, (int(t.item()), expect)
+
+    # migration on GPU shards: move an agent with pending traffic
+    mover = "agent1"
+    src = [a for a in own if a != mover][0]
+    svc.send_message(src, "pending for mover", receiver_id=mover)
+    svc.tick()
+    if rank == 0:
+        svc.migrate_agent(mover, (svc.owner_rank(mover) + 1) %% world)
+    svc.tick(); svc.tick()
+    if svc.is_local(mover):
+        msgs = svc.receive_messages(mover, max_messages=10, timeout=0)
+        assert any(m.content == "pending for mover" for m in msgs), [
+            m.content for m in msgs]
+        print(json.dumps({"ok": True, "delivered": int(t.item())}),
+              flush=True)
+    svc.close()
+    dist.destroy_process_group()
+    """
+) % str(REPO)
+
+
+def test_distributed_service_world2_gpu_shards(tmp_path):
+    """DistributedSwarmsDB over real GPU engines (2 ranks, one device,
+    gloo exchange): cross-rank delivery + live migration on device-
+    resident shards."""
+    import torch
+
+    assert torch.cuda.is_available()
+    script = tmp_path / "svc_worker.py"
+    script.write_text(WORKER_SVC)
+    env = dict(os.environ)
+    env.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr=127.0.0.1", "--master-port=29533",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=420, env=env,
+        cwd=str(REPO),
+    )
+    out = proc.stdout + proc.stderr
+    assert proc.returncode == 0, out
+    assert '"ok": true' in out.lower()
