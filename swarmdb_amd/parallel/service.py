@@ -288,16 +288,18 @@ class DistributedSwarmsDB(SwarmsDB):
         )
         # the single-process facade spills oversized content to a host
         # overflow store; across ranks the payload must fit a device slot
-        # (the owner rank has no host copy) — fail loudly at send time
-        approx = len(encode_content(content)[0]) + len(
-            encode_extras(msg.id, msg.metadata, vis)
-        )
+        # (the owner rank has no host copy) — fail loudly at send time.
+        # Encode ONCE here and carry the bytes to materialization.
+        content_b, is_json = encode_content(content)
+        extras_b = encode_extras(msg.id, msg.metadata, vis)
+        approx = len(content_b) + len(extras_b)
         if approx > int(self.config.slot_bytes):
             raise ValueError(
                 f"content ({approx} B) exceeds the distributed slot "
                 f"capacity ({self.config.slot_bytes} B); raise "
                 "SWARMQ_SLOT_BYTES"
             )
+        msg._wire = (content_b, is_json, extras_b)  # type: ignore[attr-defined]
         with self._lock:
             self._out_msgs.append(msg)
             self._pending_meta[msg.id] = msg
@@ -310,8 +312,12 @@ class DistributedSwarmsDB(SwarmsDB):
                           bitmap_idx: Optional[int] = None) -> None:
         """Turn a queued Message into a routed record (after control ops
         of the tick are applied, so indices/bitmaps exist)."""
-        content_b, is_json = encode_content(msg.content)
-        extras_b = encode_extras(msg.id, msg.metadata, msg.visible_to)
+        wire = getattr(msg, "_wire", None)
+        if wire is not None:
+            content_b, is_json, extras_b = wire
+        else:
+            content_b, is_json = encode_content(msg.content)
+            extras_b = encode_extras(msg.id, msg.metadata, msg.visible_to)
         payload = content_b + extras_b
         pad = (-len(payload)) % 16
         payload += b"\x00" * pad

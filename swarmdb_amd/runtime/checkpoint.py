@@ -290,6 +290,14 @@ def load_checkpoint(db, path: Union[str, Path],
                     with_deltas: bool = True) -> int:
     """Replay a base checkpoint (and its delta chain) into ``db``.
     Returns total records loaded."""
+    if getattr(db, "world", 1) > 1:
+        # a distributed service queues registrations for the next tick,
+        # so the replay's index remap can't be built synchronously;
+        # restore each shard into a single-rank facade instead
+        raise NotImplementedError(
+            "load_checkpoint into a DistributedSwarmsDB is not "
+            "supported; restore the shard into a single-rank SwarmsDB"
+        )
     path = Path(path)
     total = 0
     for p in [path] + (
