@@ -450,8 +450,10 @@ def main() -> int:
     stop_busy = True
 
     # latency phase: same steady state, sampling on, UNTIMED (every
-    # rank runs the same count — the distributed step has collectives)
-    lat_steps = max(4, args.steps // 4)
+    # rank runs the same count — the distributed step has collectives);
+    # capped at 64 steps (8k samples) so soak-length runs don't spend
+    # minutes sampling
+    lat_steps = max(4, min(args.steps // 4, 64))
     sampling["on"] = True
     lat_sent = lat_recv = 0
     for _ in range(lat_steps):
