@@ -61,6 +61,21 @@ def main() -> None:
         print(f"stats: total={stats['total_messages']} "
               f"by_type={stats['messages_by_type']} "
               f"by_status={stats['messages_by_status']}")
+
+        # express lane: persistent-kernel single-message latency plane
+        # on GPU (9.7 us p50 measured), in-process double on CPU
+        db.express_start(["agent1", "agent2"])
+        db.express_send("agent1", "agent2", "urgent ping")
+        sender, payload = db.express_recv("agent2", timeout_us=1e6)
+        print(f"express: {sender} -> {payload!r}")
+        db.express_stop()
+
+        # binary checkpoint: full snapshot + append-only delta
+        base = db.save_checkpoint()
+        db.send_message("agent1", "after the checkpoint",
+                        receiver_id="agent2")
+        _, n = db.save_checkpoint_delta()
+        print(f"checkpoint: {base} (+{n} delta records)")
     print("closed (history saved)")
 
 
