@@ -56,6 +56,19 @@ def main() -> None:
     stats = svc.get_stats()
     if rank == 0:
         print(f"[rank {rank}] node stats: {stats['messages_by_status']}")
+
+    # live re-sharding: move agent0 to the other rank, pending traffic
+    # re-homed through the exchange
+    world = dist.get_world_size()
+    if world > 1:
+        target = (svc.owner_rank("agent0") + 1) % world
+        if rank == 0:
+            svc.migrate_agent("agent0", target)
+        svc.tick()  # op applies + old owner drains
+        svc.tick()  # re-homed records deliver
+        if svc.is_local("agent0"):
+            print(f"[rank {rank}] agent0 migrated here "
+                  f"(owner={svc.owner_rank('agent0')})")
     svc.config.auto_save = False
     dist.destroy_process_group()
 
