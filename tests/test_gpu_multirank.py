@@ -319,3 +319,34 @@ def test_distributed_service_world2_gpu_shards(tmp_path):
     out = proc.stdout + proc.stderr
     assert proc.returncode == 0, out
     assert '"ok": true' in out.lower()
+
+
+def test_bench_distributed_gloo_gpu_engines(tmp_path):
+    """bench.py's distributed branch with GPU engines and the host-path
+    (gloo) exchange — the exact fallback the 8-GPU scaling bench takes
+    if the RCCL GPU-direct path ever fails, proven on hardware."""
+    import torch
+
+    assert torch.cuda.is_available()
+    env = dict(os.environ)
+    env.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    env["SWARMDB_BENCH_BACKEND"] = "gloo"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr=127.0.0.1", "--master-port=29534",
+            str(REPO / "bench.py"), "--gpus", "2", "--steps", "4",
+            "--warmup", "1", "--agents", "64", "--batch", "512",
+        ],
+        capture_output=True, text=True, timeout=420, env=env,
+        cwd=str(REPO),
+    )
+    out = proc.stdout + proc.stderr
+    assert proc.returncode == 0, out
+    line = [ln for ln in proc.stdout.splitlines()
+            if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["config"]["engine"] == "GpuEngine"
+    assert d["n_gpus"] == 2  # world size (both ranks on one device here)
+    assert d["value"] > 0 and d["p50_latency_ms"] > 0
