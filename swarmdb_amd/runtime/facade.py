@@ -794,17 +794,23 @@ class SwarmsDB:
         return self._messages_from_seqs(seqs)
 
     def get_conversation(
-        self, agent1_id: str, agent2_id: str, limit: int = 100
+        self, agent1_id: str, agent2_id: str, limit: int = 100,
+        sort: bool = False,
     ) -> List[Message]:
         """Two half-limit queries concatenated, not interleaved (reference
-        swarmdb/ main.py:783-808 — kept)."""
+        swarmdb/ main.py:783-808 — kept). ``sort=True``
+        opts into chronological interleaving (the behavior the reference
+        presumably intended)."""
         a_to_b = self.query_messages(
             sender_id=agent1_id, receiver_id=agent2_id, limit=limit // 2
         )
         b_to_a = self.query_messages(
             sender_id=agent2_id, receiver_id=agent1_id, limit=limit // 2
         )
-        return a_to_b + b_to_a
+        out = a_to_b + b_to_a
+        if sort:
+            out.sort(key=lambda m: m.timestamp)
+        return out
 
     def get_unread_message_count(self, agent_id: str) -> int:
         """Inbox entries still in DELIVERED state (reference swarmdb/

@@ -518,3 +518,22 @@ def test_express_lane_cpu_double(tmp_path):
     with pytest.raises(RuntimeError):
         db.express_send("a", "b", "nope")
     db.close()
+
+
+def test_get_conversation_sort_flag(tmp_path):
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path),
+                      auto_save=False, max_agents=64)
+    db = SwarmsDB(config=cfg)
+    db.send_message("a", "1", receiver_id="b")
+    time.sleep(0.002)
+    db.send_message("b", "2", receiver_id="a")
+    time.sleep(0.002)
+    db.send_message("a", "3", receiver_id="b")
+    # default keeps the reference's unmerged concatenation
+    plain = [m.content for m in db.get_conversation("a", "b", limit=10)]
+    assert plain == ["3", "1", "2"]
+    # sort=True interleaves chronologically
+    merged = [m.content for m in db.get_conversation("a", "b", limit=10,
+                                                     sort=True)]
+    assert merged == ["1", "2", "3"]
+    db.close()
