@@ -14,7 +14,7 @@ from .core.config import QueueConfig
 from .core.message import Message, MessagePriority, MessageStatus, MessageType
 from .runtime.facade import SwarmsDB
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = [
     "SwarmsDB",
