@@ -207,7 +207,7 @@ def main() -> int:
     # costs ~6% throughput, which must not contaminate the timed
     # region); the latencies are still real per-message measurements.
     sampling = {"on": False}
-    lat_samples: list = []  # (t_recv, seqs ndarray)
+    lat_deltas: list = []  # per-message (t_recv - submission) seconds
 
     def keep_sample(t_recv: float, seqs, counts=None, K=None) -> None:
         if seqs is None or len(seqs) == 0:
@@ -222,7 +222,14 @@ def main() -> int:
                 return
         take = min(128, len(seqs))
         idx = rng.choice(len(seqs), take, replace=False)
-        lat_samples.append((t_recv, seqs[idx].copy()))
+        # fetch the headers NOW (the phase is untimed): deferring the
+        # fetch lets the slot ring evict/reuse sampled seqs first
+        hdrs, _ = engine.fetch(seqs[idx].copy())
+        ts = np.asarray(hdrs["timestamp"], dtype=np.float64)
+        d = t_recv - ts
+        d = d[(ts > 0) & (d >= 0) & (d < 60.0)]
+        if len(d):
+            lat_deltas.append(d)
 
     def barrier_sync():
         if dist_on:
@@ -454,8 +461,18 @@ def main() -> int:
     # capped at 64 steps (8k samples) so soak-length runs don't spend
     # minutes sampling
     lat_steps = max(4, min(args.steps // 4, 64))
-    sampling["on"] = True
+    # two unsampled flush steps first: the pipelined path prefetched
+    # (and timestamp-stamped) one batch BEFORE the end-of-loop barrier,
+    # so its age includes the barrier/bookkeeping gap — draining it
+    # unsampled keeps that harness artifact out of the percentiles
     lat_sent = lat_recv = 0
+    for _ in range(2):
+        before_s, before_r = sent_total, recv_total
+        step(it)
+        it += 1
+        lat_sent += sent_total - before_s
+        lat_recv += recv_total - before_r
+    sampling["on"] = True
     for _ in range(lat_steps):
         before_s, before_r = sent_total, recv_total
         step(it)
@@ -499,24 +516,14 @@ def main() -> int:
     ms_per_step = elapsed / args.steps * 1000.0
 
     # per-message latency: receive wall time minus the submission stamp
-    # carried in each sampled message's header (computed OUTSIDE the
-    # timed region; evicted samples — zeroed headers — are filtered)
+    # carried in each sampled message's header
     p50_ms = p99_ms = None
     n_lat = 0
-    if lat_samples:
-        deltas = []
-        for t_recv, sample_seqs in lat_samples:
-            hdrs, _ = engine.fetch(sample_seqs)
-            ts = np.asarray(hdrs["timestamp"], dtype=np.float64)
-            d = t_recv - ts
-            d = d[(ts > 0) & (d >= 0) & (d < 60.0)]
-            if len(d):
-                deltas.append(d)
-        if deltas:
-            alld = np.concatenate(deltas)
-            n_lat = int(len(alld))
-            p50_ms = float(np.median(alld) * 1000.0)
-            p99_ms = float(np.percentile(alld, 99) * 1000.0)
+    if lat_deltas:
+        alld = np.concatenate(lat_deltas)
+        n_lat = int(len(alld))
+        p50_ms = float(np.median(alld) * 1000.0)
+        p99_ms = float(np.percentile(alld, 99) * 1000.0)
     if p50_ms is None:
         # no samples (zero deliveries in the latency phase): fall back to
         # tick-duration bound and say so via n_lat_samples = 0
