@@ -460,11 +460,14 @@ def main() -> int:
     # rank runs the same count — the distributed step has collectives);
     # capped at 64 steps (8k samples) so soak-length runs don't spend
     # minutes sampling
-    lat_steps = max(4, min(args.steps // 4, 64))
-    # two unsampled flush steps first: the pipelined path prefetched
-    # (and timestamp-stamped) one batch BEFORE the end-of-loop barrier,
-    # so its age includes the barrier/bookkeeping gap — draining it
-    # unsampled keeps that harness artifact out of the percentiles
+    lat_steps = max(8, min(args.steps // 2, 128))
+    # Alternate sampled/unsampled steps: a sampled step's bookkeeping
+    # (drain + header fetch) runs AFTER its own receive stamp but ages
+    # the NEXT step's messages — with every step sampled that observer
+    # overhead compounds into the percentiles. Alternation means every
+    # sampled step measures messages stamped during a clean step. Two
+    # unsampled flush steps first drain the batch that was prefetched
+    # (and stamped) before the end-of-loop barrier.
     lat_sent = lat_recv = 0
     for _ in range(2):
         before_s, before_r = sent_total, recv_total
@@ -472,8 +475,8 @@ def main() -> int:
         it += 1
         lat_sent += sent_total - before_s
         lat_recv += recv_total - before_r
-    sampling["on"] = True
-    for _ in range(lat_steps):
+    for j in range(lat_steps):
+        sampling["on"] = j % 2 == 1
         before_s, before_r = sent_total, recv_total
         step(it)
         it += 1
