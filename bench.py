@@ -475,14 +475,20 @@ def main() -> int:
         it += 1
         lat_sent += sent_total - before_s
         lat_recv += recv_total - before_r
+    phase_times = []
     for j in range(lat_steps):
         sampling["on"] = j % 2 == 1
         before_s, before_r = sent_total, recv_total
+        ps = time.perf_counter()
         step(it)
+        phase_times.append(time.perf_counter() - ps)
         it += 1
         lat_sent += sent_total - before_s
         lat_recv += recv_total - before_r
     sampling["on"] = False
+    if args.dump_steps and rank == 0:
+        print("phase_times_ms:",
+              [round(t * 1000, 3) for t in phase_times])
     sent_total -= lat_sent  # latency phase is outside the timed totals
     recv_total -= lat_recv
 
