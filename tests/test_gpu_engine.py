@@ -924,3 +924,42 @@ def test_binary_checkpoint_roundtrip_gpu(tmp_path):
     ]
     db.close()
     db2.close()
+
+
+def test_doorbell_burst_ring_wrap():
+    """Express-lane stress: a 50k-message burst wraps the 256-entry
+    submit ring ~200 times and the 64-entry delivery ring as the
+    consumer lags; ordering and flow control must hold throughout."""
+    from swarmdb_amd import _swarmq
+
+    db = _swarmq.DoorbellQueue(slot_bytes=256, sub_cap=256, n_agents=4,
+                               ring_cap=64, device=0)
+    db.start(60.0)
+    try:
+        n = 50_000
+        got = 0
+        next_expect = 0
+        import struct
+        for i in range(n):
+            db.send(receiver=1, sender=0,
+                    payload=struct.pack("<I", i).ljust(32, b"."))
+            # drain opportunistically so the delivery ring never drops
+            while True:
+                m = db.try_recv(1)
+                if m is None:
+                    break
+                val = struct.unpack("<I", bytes(m[1])[:4])[0]
+                assert val == next_expect, (val, next_expect)
+                next_expect += 1
+                got += 1
+        while got < n:
+            m = db.recv_spin(1, timeout_us=2e6)
+            assert m is not None, f"lost after {got}/{n}"
+            val = struct.unpack("<I", bytes(m[1])[:4])[0]
+            assert val == next_expect, (val, next_expect)
+            next_expect += 1
+            got += 1
+        assert db.consumed() == n
+    finally:
+        db.stop()
+        db.release()
