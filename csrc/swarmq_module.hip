@@ -2005,7 +2005,10 @@ struct DoorGeom {
 __global__ void k_doorbell(volatile ull *ctrl, const Rec *__restrict__ sub_recs,
                            const u8 *__restrict__ sub_pay,
                            Rec *__restrict__ del_recs, u8 *__restrict__ del_pay,
-                           volatile ull *del_count, ull *__restrict__ del_next,
+                           volatile ull *del_count,
+                           volatile const ull *del_rpos,
+                           ull *__restrict__ del_next,
+                           ull *__restrict__ rpos_cache,
                            DoorGeom dg, unsigned long long max_cycles) {
   const int lane = threadIdx.x & 63;
   const unsigned long long t0 = __builtin_amdgcn_s_memrealtime();
@@ -2039,6 +2042,31 @@ __global__ void k_doorbell(volatile ull *ctrl, const Rec *__restrict__ sub_recs,
       const u32 plen = (u32)recv_len;
       if (recv < dg.n_agents) {
         const u64 pos = del_next[recv];
+        // delivery-ring back-pressure: never overwrite an entry the
+        // host hasn't consumed. The host's read cursor lives in pinned
+        // memory (PCIe-latency read), so a device-side cache
+        // (rpos_cache) keeps the hot path free of host reads: refresh
+        // only when the ring LOOKS full against the cache. A slow
+        // consumer stalls the lane (the express plane assumes prompt
+        // consumers); the wall-clock budget still bounds the kernel.
+        u32 ok = 1;
+        if (lane == 0 && pos - rpos_cache[recv] >= (u64)dg.ring_cap) {
+          for (;;) {
+            const u64 rp = del_rpos[recv];
+            rpos_cache[recv] = rp;
+            if (pos - rp < (u64)dg.ring_cap)
+              break;
+            if (ctrl[DB_STOP] ||
+                __builtin_amdgcn_s_memrealtime() - t0 > max_cycles) {
+              ok = 0;
+              break;
+            }
+            __builtin_amdgcn_s_sleep(32);
+          }
+        }
+        ok = __shfl(ok, 0, 64);
+        if (!ok)
+          break; // stop/timeout while full: exit without overwriting
         const u64 slot = (u64)recv * dg.ring_cap + (pos % dg.ring_cap);
         // payload copy: 64 lanes x 16 B per round
         const uint4 *src = reinterpret_cast<const uint4 *>(
@@ -2096,11 +2124,17 @@ public:
                             (size_t)n_agents * ring_cap * slot_bytes, flags));
     HIP_CHECK(hipHostMalloc((void **)&h_del_count_,
                             (size_t)n_agents * sizeof(ull), flags));
+    HIP_CHECK(hipHostMalloc((void **)&h_del_rpos_,
+                            (size_t)n_agents * sizeof(ull), flags));
     std::memset((void *)h_ctrl_, 0, DB_NWORDS * sizeof(ull));
     std::memset((void *)h_del_count_, 0, (size_t)n_agents * sizeof(ull));
-    // device-private per-agent delivery cursor (only the kernel touches it)
+    std::memset((void *)h_del_rpos_, 0, (size_t)n_agents * sizeof(ull));
+    // device-private per-agent delivery cursor + host-read-pos cache
+    // (only the kernel touches these)
     HIP_CHECK(hipMalloc(&d_del_next_, (size_t)n_agents * sizeof(ull)));
     HIP_CHECK(hipMemset(d_del_next_, 0, (size_t)n_agents * sizeof(ull)));
+    HIP_CHECK(hipMalloc(&d_del_rcache_, (size_t)n_agents * sizeof(ull)));
+    HIP_CHECK(hipMemset(d_del_rcache_, 0, (size_t)n_agents * sizeof(ull)));
     read_pos_.assign(n_agents, 0);
     HIP_CHECK(hipHostGetDevicePointer((void **)&m_ctrl_, (void *)h_ctrl_, 0));
     HIP_CHECK(hipHostGetDevicePointer((void **)&m_sub_recs_, h_sub_recs_, 0));
@@ -2109,6 +2143,8 @@ public:
     HIP_CHECK(hipHostGetDevicePointer((void **)&m_del_pay_, h_del_pay_, 0));
     HIP_CHECK(
         hipHostGetDevicePointer((void **)&m_del_count_, (void *)h_del_count_, 0));
+    HIP_CHECK(
+        hipHostGetDevicePointer((void **)&m_del_rpos_, (void *)h_del_rpos_, 0));
   }
 
   ~DoorbellQueue() { release(); }
@@ -2121,6 +2157,8 @@ public:
       stop();
     (void)hipStreamDestroy(stream_);
     (void)hipFree(d_del_next_);
+    (void)hipFree(d_del_rcache_);
+    (void)hipHostFree((void *)h_del_rpos_);
     (void)hipHostFree((void *)h_ctrl_);
     (void)hipHostFree(h_sub_recs_);
     (void)hipHostFree(h_sub_pay_);
@@ -2140,7 +2178,8 @@ public:
         (unsigned long long)(max_seconds * 100.0e6);
     hipLaunchKernelGGL(k_doorbell, dim3(1), dim3(64), 0, stream_, m_ctrl_,
                        m_sub_recs_, m_sub_pay_, m_del_recs_, m_del_pay_,
-                       m_del_count_, d_del_next_, dg_, max_cycles);
+                       m_del_count_, m_del_rpos_, d_del_next_,
+                       d_del_rcache_, dg_, max_cycles);
     HIP_CHECK(hipGetLastError());
     running_ = true;
   }
@@ -2192,7 +2231,9 @@ public:
   }
 
   // Non-blocking poll of an agent's delivery ring. Returns
-  // (sender, payload bytes) or None.
+  // (sender, payload bytes) or None. The read position is published
+  // back to the kernel (delivery-ring back-pressure): entries are
+  // never overwritten before the host consumed them.
   py::object try_recv(u32 agent) {
     if (agent >= dg_.n_agents)
       throw std::out_of_range("agent outside the express agent set");
@@ -2200,8 +2241,6 @@ public:
     u64 &rp = read_pos_[agent];
     if (have == rp)
       return py::none();
-    if (have - rp > dg_.ring_cap) // overwritten: skip to the oldest intact
-      rp = have - dg_.ring_cap;
     std::atomic_thread_fence(std::memory_order_seq_cst);
     const u64 slot = (u64)agent * dg_.ring_cap + (rp % dg_.ring_cap);
     const Rec r = h_del_recs_[slot];
@@ -2209,6 +2248,8 @@ public:
         reinterpret_cast<const char *>(h_del_pay_ + slot * dg_.slot_bytes),
         r.payload_len);
     ++rp;
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    h_del_rpos_[agent] = rp; // release the slot to the kernel
     return py::make_tuple(r.sender, pay);
   }
 
@@ -2256,7 +2297,9 @@ private:
   Rec *h_del_recs_{};
   u8 *h_del_pay_{};
   volatile ull *h_del_count_{};
+  volatile ull *h_del_rpos_{};
   ull *d_del_next_{};
+  ull *d_del_rcache_{};
   std::vector<u64> read_pos_;
   // device-visible mappings of the pinned blocks
   volatile ull *m_ctrl_{};
@@ -2265,6 +2308,7 @@ private:
   Rec *m_del_recs_{};
   u8 *m_del_pay_{};
   volatile ull *m_del_count_{};
+  volatile ull *m_del_rpos_{};
 };
 
 // ---------------------------------------------------------------------------
