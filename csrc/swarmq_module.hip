@@ -2210,10 +2210,22 @@ public:
     if (receiver >= dg_.n_agents)
       throw std::out_of_range("receiver outside the express agent set");
     const u64 h = sub_head_;
-    // bounded wait for ring space (the kernel is normally far ahead)
-    for (int spins = 0; h - h_ctrl_[DB_CONSUMED] >= dg_.sub_cap; ++spins) {
-      if (spins > 50'000'000)
-        throw std::runtime_error("doorbell ring stalled (kernel dead?)");
+    // bounded wait for ring space (the kernel is normally far ahead).
+    // GIL released while spinning: the kernel may be stalled on a FULL
+    // delivery ring, which only a consumer thread can drain — holding
+    // the GIL here would deadlock multithreaded producers/consumers.
+    if (h - h_ctrl_[DB_CONSUMED] >= dg_.sub_cap) {
+      py::gil_scoped_release nogil;
+      const auto t0 = std::chrono::steady_clock::now();
+      while (h - h_ctrl_[DB_CONSUMED] >= dg_.sub_cap) {
+        std::this_thread::yield();
+        if (std::chrono::steady_clock::now() - t0 >
+            std::chrono::seconds(5))
+          throw std::runtime_error(
+              "doorbell submit ring stalled for 5 s — kernel dead, or "
+              "every consumer stopped draining (the delivery-ring "
+              "back-pressure holds the kernel until try_recv runs)");
+      }
     }
     const u32 i = (u32)(h % dg_.sub_cap);
     Rec r{};

@@ -940,18 +940,28 @@ def test_doorbell_burst_ring_wrap():
         got = 0
         next_expect = 0
         import struct
-        for i in range(n):
-            db.send(receiver=1, sender=0,
-                    payload=struct.pack("<I", i).ljust(32, b"."))
-            # drain opportunistically so the delivery ring never drops
+
+        def drain():
+            nonlocal got, next_expect
             while True:
                 m = db.try_recv(1)
                 if m is None:
-                    break
+                    return
                 val = struct.unpack("<I", bytes(m[1])[:4])[0]
                 assert val == next_expect, (val, next_expect)
                 next_expect += 1
                 got += 1
+
+        for i in range(n):
+            # bounded in-flight: a single-threaded producer+consumer
+            # must drain before the delivery ring fills (the kernel
+            # back-pressures rather than overwrite; a blocked send
+            # can't be un-blocked by the SAME thread's draining)
+            if i - got >= 48:
+                drain()
+            db.send(receiver=1, sender=0,
+                    payload=struct.pack("<I", i).ljust(32, b"."))
+            drain()
         while got < n:
             m = db.recv_spin(1, timeout_us=2e6)
             assert m is not None, f"lost after {got}/{n}"
