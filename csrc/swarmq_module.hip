@@ -2199,6 +2199,11 @@ public:
   bool running() const { return running_; }
   bool exited() const { return h_ctrl_[DB_EXITED] != 0; }
   u64 consumed() const { return h_ctrl_[DB_CONSUMED]; }
+  u64 head() const { return sub_head_; }
+  u64 delivered_count(u32 agent) const {
+    return ((volatile ull *)h_del_count_)[agent];
+  }
+  u64 read_pos(u32 agent) const { return read_pos_[agent]; }
 
   // Submit one message to the express lane. Spins (bounded) if the ring
   // is full. Returns the express sequence number.
@@ -2397,6 +2402,9 @@ PYBIND11_MODULE(_swarmq, m) {
       .def("running", &DoorbellQueue::running)
       .def("exited", &DoorbellQueue::exited)
       .def("consumed", &DoorbellQueue::consumed)
+      .def("head", &DoorbellQueue::head)
+      .def("delivered_count", &DoorbellQueue::delivered_count)
+      .def("read_pos", &DoorbellQueue::read_pos)
       .def("send", &DoorbellQueue::send, py::arg("receiver"),
            py::arg("sender"), py::arg("payload"))
       .def("try_recv", &DoorbellQueue::try_recv)

@@ -954,11 +954,16 @@ def test_doorbell_burst_ring_wrap():
 
         for i in range(n):
             # bounded in-flight: a single-threaded producer+consumer
-            # must drain before the delivery ring fills (the kernel
-            # back-pressures rather than overwrite; a blocked send
-            # can't be un-blocked by the SAME thread's draining)
-            if i - got >= 48:
-                drain()
+            # must WAIT for drains before the delivery ring fills (the
+            # kernel back-pressures rather than overwrite; a blocked
+            # send can't be un-blocked by the SAME thread's draining)
+            while i - got >= 48:
+                m = db.recv_spin(1, timeout_us=2e6)
+                assert m is not None, f"kernel stalled at {got}/{i}"
+                val = struct.unpack("<I", bytes(m[1])[:4])[0]
+                assert val == next_expect, (val, next_expect)
+                next_expect += 1
+                got += 1
             db.send(receiver=1, sender=0,
                     payload=struct.pack("<I", i).ljust(32, b"."))
             drain()
