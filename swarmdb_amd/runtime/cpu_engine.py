@@ -37,6 +37,12 @@ from .engine import (
 
 _GROW = 2
 
+# fetch output layout: REC fields + status + seq (precomputed — building
+# it per call costs a dtype introspection per fetch)
+_FETCH_DTYPE = np.dtype(
+    REC_DTYPE.descr + [("status", np.uint8), ("seq", np.uint64)]
+)
+
 
 class _U64Ring:
     """Growable append-only u64 vector (inbox log)."""
@@ -317,10 +323,7 @@ class CpuEngine(Engine):
                 bytes(mv[int(o) : int(o) + int(l)])
                 for o, l in zip(self._pay_off[seqs], self._pay_len[seqs])
             ]
-            out = np.zeros(
-                len(seqs),
-                dtype=REC_DTYPE.descr + [("status", np.uint8), ("seq", np.uint64)],
-            )
+            out = np.zeros(len(seqs), dtype=_FETCH_DTYPE)
             for name in REC_DTYPE.names:
                 out[name] = hdr[name]
             out["status"] = status
@@ -334,10 +337,7 @@ class CpuEngine(Engine):
             seqs = np.asarray(seqs, dtype=np.int64)
             n = len(seqs)
             stride = int(self.cfg.slot_bytes)
-            out = np.zeros(
-                n,
-                dtype=REC_DTYPE.descr + [("status", np.uint8), ("seq", np.uint64)],
-            )
+            out = np.zeros(n, dtype=_FETCH_DTYPE)
             h = self._hdr[seqs]
             for name in REC_DTYPE.names:
                 out[name] = h[name]
