@@ -108,3 +108,68 @@ def test_gateway_world2(tmp_path):
     )
     assert proc.returncode == 0, proc.stdout + proc.stderr
     assert '"ok": true' in proc.stdout.lower()
+
+
+WORKER_SERVE = textwrap.dedent(
+    """
+    import json, sys
+    sys.path.insert(0, %r)
+    import torch.distributed as dist
+    from fastapi.testclient import TestClient
+
+    from swarmdb_amd.api.serve_distributed import build
+
+    svc, app, port = build()
+    rank = dist.get_rank()
+    assert port == 8000 + rank
+    client = TestClient(app, follow_redirects=False)
+    assert client.get("/health").json()["status"] == "healthy"
+    tok = client.post("/auth/token",
+                      json={"username": f"w{rank}", "password": "x"}
+                      ).json()["access_token"]
+    h = {"Authorization": "Bearer " + tok}
+    client.post("/agents/register", headers=h,
+                json={"agent_id": f"w{rank}"})
+    # the background ticker applies the registration within a few ticks
+    import time
+    for _ in range(200):
+        if f"w{rank}" in svc._agent_idx:
+            break
+        time.sleep(0.01)
+    assert f"w{rank}" in svc._agent_idx
+    # wait until BOTH ranks' registrations propagated
+    for _ in range(300):
+        if len(svc._agent_idx) == 2:
+            break
+        time.sleep(0.01)
+    assert len(svc._agent_idx) == 2, svc._agent_idx
+    svc.close()
+    if rank == 0:
+        print(json.dumps({"ok": True}))
+    dist.destroy_process_group()
+    """
+) % str(REPO)
+
+
+def test_serve_distributed_glue_world2(tmp_path):
+    """The per-rank server entry (`swarmdb_amd.api.serve_distributed`):
+    service + gateway app + background ticker built per rank; REST
+    registrations propagate across ranks through the ticker."""
+    script = tmp_path / "serve_worker.py"
+    script.write_text(WORKER_SERVE)
+    env = dict(os.environ)
+    env.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    env["SWARMDB_BASE_PORT"] = "8000"
+    env["MESSAGE_HISTORY_DIR"] = str(tmp_path / "hist")
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr=127.0.0.1", "--master-port=29543",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=240, env=env,
+        cwd=str(REPO),
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert '"ok": true' in proc.stdout.lower()
