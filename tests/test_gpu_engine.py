@@ -978,3 +978,22 @@ def test_doorbell_burst_ring_wrap():
     finally:
         db.stop()
         db.release()
+
+
+def test_recv_event_log_prunes_by_age(gpu_engine):
+    """processing_rate probe log prunes by AGE (round-1 advisor: the
+    fixed-count prune truncated the 60s window under fast polling)."""
+    eng = gpu_engine
+    eng.register_agent(0)
+    now = time.time()
+    # synthetic history: stale entries beyond the retention window plus
+    # fresh ones inside the probe window
+    stale = [(now - 700.0, np.array([0], dtype=np.uint32),
+              np.array([5], dtype=np.int64)) for _ in range(2000)]
+    fresh = [(now - 10.0, np.array([0], dtype=np.uint32),
+              np.array([3], dtype=np.int64)) for _ in range(50)]
+    eng._recv_events = stale + fresh
+    eng.receive_many(np.array([0], dtype=np.uint32), 10)
+    # all 2000 stale entries gone; the fresh window intact
+    assert len(eng._recv_events) <= 52
+    assert eng.recv_rate_window(0, 60.0) == 50 * 3
