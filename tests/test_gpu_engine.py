@@ -993,7 +993,11 @@ def test_recv_event_log_prunes_by_age(gpu_engine):
     fresh = [(now - 10.0, np.array([0], dtype=np.uint32),
               np.array([3], dtype=np.int64)) for _ in range(50)]
     eng._recv_events = stale + fresh
+    # the bookkeeping (append + age prune) runs on a delivering poll
+    recs, payload = make_batch(np.random.default_rng(0), 4, 1)
+    recs["receiver"] = 0
+    eng.enqueue_batch(recs, payload)
     eng.receive_many(np.array([0], dtype=np.uint32), 10)
-    # all 2000 stale entries gone; the fresh window intact
+    # all 2000 stale entries gone; the fresh window + this poll intact
     assert len(eng._recv_events) <= 52
-    assert eng.recv_rate_window(0, 60.0) == 50 * 3
+    assert eng.recv_rate_window(0, 60.0) == 50 * 3 + 4
