@@ -52,6 +52,12 @@ class QueueConfig:
     log_file: Optional[str] = None       # rotating file log (reference
     #                                      loguru sink, swarmdb/ main.py:170-189)
     save_dir: str = "message_history"
+    history_indent: int = 2              # reference snapshot format uses
+    #                                      indent=2 (main.py:886); 0 opts
+    #                                      into compact JSON, which uses
+    #                                      Python's C encoder (~6x faster
+    #                                      serialization — same structure,
+    #                                      different whitespace)
     auto_save: bool = True
     save_interval: float = 300.0
     max_messages_per_file: int = 10000
@@ -82,6 +88,7 @@ class QueueConfig:
             inbox_capacity=int(env.get("SWARMQ_INBOX_CAPACITY", str(1 << 16))),
             staging_batch=int(env.get("SWARMQ_STAGING_BATCH", "16384")),
             num_bitmaps=int(env.get("SWARMQ_NUM_BITMAPS", "4096")),
+            history_indent=int(env.get("SWARMDB_HISTORY_INDENT", "2")),
             device_index=int(env.get("SWARMQ_DEVICE", "0")),
         )
         kw.update(overrides)

@@ -1130,8 +1130,11 @@ class SwarmsDB:
         self.save_dir.mkdir(parents=True, exist_ok=True)
         ts = datetime.now().strftime("%Y%m%d_%H%M%S")
         path = self.save_dir / f"message_history_{ts}_{history['message_count']}.json"
+        # indent=2 is the reference's on-disk format; indent 0 (config)
+        # selects compact JSON, which runs on the C encoder (~6x faster)
+        indent = self.config.history_indent or None
         with open(path, "w") as f:
-            json.dump(history, f, indent=2)
+            json.dump(history, f, indent=indent)
         with self._lock:
             sidecar = {
                 "agent_groups": self.metadata.get("agent_groups", {}),
@@ -1139,7 +1142,7 @@ class SwarmsDB:
                 "agent_metadata": self.agent_metadata,
             }
         with open(self.save_dir / f"metadata_{ts}.json", "w") as f:
-            json.dump(sidecar, f, indent=2)
+            json.dump(sidecar, f, indent=indent)
         self.last_save_time = time.time()
         logger.info("saved history to %s", path)
         return str(path)
@@ -1286,7 +1289,7 @@ class SwarmsDB:
         archive_dir.mkdir(parents=True, exist_ok=True)
         path = archive_dir / f"archive_{int(time.time())}.json"
         with open(path, "w") as f:
-            json.dump(archive, f, indent=2)
+            json.dump(archive, f, indent=self.config.history_indent or None)
         logger.info("flushed %d old messages to %s", count, path)
         return count
 
