@@ -205,3 +205,37 @@ def test_binary_checkpoint_roundtrip(tmp_path):
     assert [m.content for m in msgs] == ["plain one"]
     db.close()
     db2.close()
+
+
+def test_binary_checkpoint_replay_into_nonempty_facade(tmp_path):
+    """The replay remaps dense agent indices: loading into a facade
+    whose index table differs (pre-existing agents in another order)
+    must still deliver to the right agents and rebuild visibility."""
+    cfg = dict(use_gpu=False, save_dir=str(tmp_path), auto_save=False,
+               max_agents=64, slot_bytes=512)
+    db = SwarmsDB(config=QueueConfig(**cfg))
+    for a in ["alice", "bob", "carol"]:
+        db.register_agent(a)  # alice=0, bob=1, carol=2
+    db.send_message("alice", "for bob", receiver_id="bob")
+    db.send_message("bob", "members only", receiver_id=None,
+                    visible_to=["alice"])
+    base = db.save_checkpoint()
+
+    db2 = SwarmsDB(config=QueueConfig(**cfg))
+    # conflicting pre-existing table: carol=0, zed=1, bob=2, alice=3
+    for a in ["carol", "zed", "bob", "alice"]:
+        db2.register_agent(a)
+    db2.send_message("zed", "pre-existing", receiver_id="bob")
+    loaded = db2.load_checkpoint(base)
+    assert loaded == 2
+    got_bob = sorted(m.content for m in db2.receive_messages("bob",
+                                                             timeout=0))
+    assert got_bob == ["for bob", "pre-existing"]
+    got_alice = [m.content for m in db2.receive_messages("alice",
+                                                         timeout=0)]
+    assert got_alice == ["members only"]
+    # the restricted message never reaches non-members
+    assert db2.receive_messages("zed", timeout=0) == []
+    assert db2.receive_messages("carol", timeout=0) == []
+    db.close()
+    db2.close()
