@@ -6,10 +6,12 @@ RCCL over xGMI on GPU, gloo on CPU for CI). Replaces the reference's
 "cross-process shared state") with:
 
 - **control plane, tick-synchronized**: agent registrations, visibility
-  bitmaps and group definitions are queued as ops and applied on EVERY
-  rank in a deterministic order each tick (``all_gather_object``), so the
-  dense agent-index table and bitmap pool stay bit-identical across
-  ranks with no coordinator;
+  bitmaps, group definitions and migrations are queued as ops and
+  applied on EVERY rank in a deterministic order each tick (a fused
+  tensor header exchange + a padded-u8 JSON all_gather only when ops
+  exist — no pickle, quiet tick = one tiny collective), so the dense
+  agent-index table, ownership table and bitmap pool stay bit-identical
+  across ranks with no coordinator;
 - **data plane**: outbound messages batch into a per-tick exchange
   routed by owner rank (dense agent index mod world — the same mapping
   the all-to-all router applies) via all-to-all; an agent's inbox lives
