@@ -406,10 +406,14 @@ def main() -> int:
         extra = int((0.15 - elapsed_w) / per) + 1
     if dist_on:
         # every rank must run the same number of steps (collectives):
-        # agree on the max
+        # agree on the max. Tensor device must match the backend (nccl
+        # reduces CUDA tensors only).
         import torch.distributed as dist
 
-        t = torch.tensor([extra], dtype=torch.int64)
+        t = torch.tensor(
+            [extra], dtype=torch.int64,
+            device="cuda" if (have_gpu and backend == "nccl") else "cpu",
+        )
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         extra = int(t.item())
     for _ in range(extra):
