@@ -321,7 +321,10 @@ def config9_distributed_service(seconds: float) -> None:
         lat.append(time.perf_counter() - s)
         ticks += 1
     elapsed = time.perf_counter() - t0
-    t = torch.tensor([float(sent), float(delivered)])
+    t = torch.tensor(
+        [float(sent), float(delivered)],
+        device="cuda" if dist.get_backend() == "nccl" else "cpu",
+    )
     dist.all_reduce(t)
     if rank == 0:
         print(json.dumps({
