@@ -556,3 +556,51 @@ def test_admin_checkpoint_routes(client):
     # non-admin rejected
     r = client.post("/admin/checkpoint", headers=auth(client, "a"))
     assert r.status_code == 403
+
+
+def test_search_route_scoping(client):
+    """/messages/search/: device-scan route; non-admins only see their
+    own traffic, admins see all."""
+    ha, hb = auth(client, "alice"), auth(client, "bob")
+    hc = auth(client, "carol")
+    client.post("/messages", headers=ha,
+                json={"receiver_id": "bob", "content": "zebra sighting"})
+    client.post("/messages", headers=hb,
+                json={"receiver_id": "alice", "content": "no zebras here"})
+    client.post("/messages", headers=hc,
+                json={"receiver_id": "carol2", "content": "zebra private"})
+    # participants see their own zebra traffic
+    r = client.get("/messages/search/?keyword=zebra", headers=ha)
+    assert r.status_code == 200
+    assert sorted(m["content"] for m in r.json()) == [
+        "no zebras here", "zebra sighting"]
+    # case-insensitive by default
+    r = client.get("/messages/search/?keyword=ZEBRA", headers=ha)
+    assert len(r.json()) == 2
+    # case-sensitive opt-in
+    r = client.get("/messages/search/?keyword=ZEBRA&case_sensitive=true",
+                   headers=ha)
+    assert r.json() == []
+    # admin sees everything
+    r = client.get("/messages/search/?keyword=zebra",
+                   headers=auth(client, "admin"))
+    assert len(r.json()) == 3
+    # carol's private message is invisible to alice
+    assert all("private" not in m["content"]
+               for m in client.get("/messages/search/?keyword=zebra",
+                                   headers=ha).json())
+
+
+def test_export_yaml_route(client):
+    ha = auth(client, "admin")
+    client.post("/messages", headers=auth(client, "a"),
+                json={"receiver_id": "b", "content": "yaml via rest"})
+    r = client.post("/admin/export_yaml", headers=ha)
+    assert r.status_code == 200
+    import yaml as _yaml
+    data = _yaml.safe_load(Path(r.json()["path"]).read_text())
+    contents = [m["content"] for m in data["messages"].values()]
+    assert "yaml via rest" in contents
+    # non-admin rejected
+    r = client.post("/admin/export_yaml", headers=auth(client, "a"))
+    assert r.status_code == 403
