@@ -1,0 +1,130 @@
+"""Property-based tests (hypothesis) for the wire/disk formats and the
+end-to-end facade delivery invariants — arbitrary unicode content,
+metadata shapes and visibility sets must round-trip exactly."""
+
+import json
+
+import pytest
+from hypothesis import HealthCheck, given, settings
+from hypothesis import strategies as st
+
+from swarmdb_amd import QueueConfig, SwarmsDB
+from swarmdb_amd.core.message import Message, MessagePriority, MessageType
+from swarmdb_amd.core.wire import (
+    decode_content,
+    decode_extras,
+    derived_id,
+    encode_content,
+    encode_extras,
+    parse_derived_id,
+)
+
+# JSON-representable content: strings, numbers, bools, None nested in
+# dicts/lists (what the reference accepts for Message.content)
+json_scalars = st.one_of(
+    st.text(max_size=40),
+    st.integers(min_value=-(2**31), max_value=2**31),
+    st.floats(allow_nan=False, allow_infinity=False, width=32),
+    st.booleans(),
+    st.none(),
+)
+json_content = st.recursive(
+    json_scalars,
+    lambda inner: st.one_of(
+        st.lists(inner, max_size=4),
+        st.dictionaries(st.text(max_size=8), inner, max_size=4),
+    ),
+    max_leaves=10,
+)
+content_strategy = st.one_of(
+    st.text(max_size=200),
+    st.lists(json_scalars, max_size=5),
+    st.dictionaries(st.text(max_size=10), json_scalars, max_size=5),
+)
+
+
+@given(content=st.one_of(st.text(max_size=500), json_content))
+def test_content_encode_decode_roundtrip(content):
+    data, is_json = encode_content(content)
+    out = decode_content(data, is_json)
+    if isinstance(content, str):
+        assert out == content
+    else:
+        # JSON round-trip semantics (tuples->lists etc. not generated)
+        assert out == json.loads(json.dumps(content))
+
+
+@given(
+    msg_id=st.text(min_size=1, max_size=64),
+    metadata=st.dictionaries(st.text(max_size=10), json_scalars, max_size=5),
+    visible_to=st.lists(st.text(max_size=16), max_size=6),
+)
+def test_extras_roundtrip(msg_id, metadata, visible_to):
+    blob = encode_extras(msg_id, metadata, visible_to)
+    out = decode_extras(blob)
+    assert out.get("id") == msg_id
+    assert out.get("metadata", {}) == metadata
+    assert out.get("visible_to", []) == visible_to
+
+
+@given(rank=st.integers(0, 2**32 - 1), seq=st.integers(0, 2**48 - 1))
+def test_derived_id_roundtrip(rank, seq):
+    mid = derived_id(rank, seq)
+    assert parse_derived_id(mid) == (rank, seq)
+    # uuid4 ids (version nibble 4) can never parse as derived (nibble 8)
+    import uuid
+
+    assert parse_derived_id(str(uuid.uuid4())) is None
+
+
+@given(
+    content=content_strategy,
+    mtype=st.sampled_from(list(MessageType)),
+    prio=st.sampled_from(list(MessagePriority)),
+    metadata=st.dictionaries(st.text(max_size=8), json_scalars, max_size=3),
+)
+def test_message_dict_roundtrip(content, mtype, prio, metadata):
+    m = Message(sender_id="s", receiver_id="r", content=content,
+                type=mtype, priority=prio, metadata=metadata)
+    d = m.to_dict()
+    # reference wire schema: enums flattened to values
+    assert d["type"] == mtype.value and d["priority"] == prio.value
+    m2 = Message.from_dict(d)
+    assert m2.content == m.content
+    assert m2.type == m.type and m2.priority == m.priority
+    assert m2.metadata == m.metadata
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(
+    content=content_strategy,
+    metadata=st.dictionaries(st.text(max_size=8), json_scalars, max_size=3),
+    vis_pick=st.integers(0, 2),
+)
+def test_facade_delivery_preserves_payload(tmp_path, content, metadata,
+                                           vis_pick):
+    """Arbitrary content/metadata survives the engine round trip and
+    visibility restriction delivers to exactly the chosen subset."""
+    cfg = QueueConfig(use_gpu=False, save_dir=str(tmp_path),
+                      auto_save=False, max_agents=64)
+    db = SwarmsDB(config=cfg)
+    agents = ["a", "b", "c"]
+    for a in agents:
+        db.register_agent(a)
+    visible_to = [["b"], ["b", "c"], None][vis_pick]
+    db.send_message("a", content, receiver_id=None, metadata=metadata,
+                    visible_to=visible_to)
+    got = {a: db.receive_messages(a, timeout=0) for a in agents}
+    # raw send_message broadcast defaults visible_to to ALL registered
+    # agents INCLUDING the sender (reference main.py:449-450; only
+    # broadcast_message excludes the sender — SURVEY.md §8.11)
+    members = set(visible_to) if visible_to else {"a", "b", "c"}
+    for a in agents:
+        if a in members:
+            assert len(got[a]) == 1, a
+            assert got[a][0].content == content
+            assert got[a][0].metadata == metadata
+        else:
+            assert got[a] == [], a
+    db.close()
